@@ -1,0 +1,152 @@
+#!/usr/bin/env python3
+"""Flagship benchmark: DWT-ResNet50 3-stream training step (Office-Home config).
+
+Measures whole-job images/sec of the full training step — forward over the
+concatenated (source, target, target_aug) batch, CE + MEC losses, backward,
+gradient all-reduce (DP>1), SGD step — on synthetic data with random-init
+weights (BASELINE.json: there is no network for datasets/checkpoints).
+
+Driver contract: `python bench.py --gpus N --steps K --warmup W`; under
+torchrun one rank per GPU (reads RANK/WORLD_SIZE/LOCAL_RANK).  Rank 0 prints
+one JSON line; `value` aggregates over all N GPUs; timing is max-over-ranks,
+bracketed by barrier + torch.cuda.synchronize on both sides.
+"""
+from __future__ import annotations
+
+import argparse
+import json
+import os
+import time
+
+import torch
+import torch.distributed as dist
+import torch.nn.functional as F
+
+
+def get_args():
+    p = argparse.ArgumentParser()
+    p.add_argument("--gpus", type=int, default=1)
+    p.add_argument("--steps", type=int, default=20)
+    p.add_argument("--warmup", type=int, default=5)
+    p.add_argument("--batch", type=int, default=64,
+                   help="per-domain per-GPU batch (total 3x this per GPU)")
+    p.add_argument("--img", type=int, default=224)
+    p.add_argument("--num_classes", type=int, default=65)
+    p.add_argument("--dtype", choices=["bfloat16", "float32"], default="bfloat16")
+    p.add_argument("--group_size", type=int, default=4)
+    p.add_argument("--whiten_mode", choices=["chol", "zca"], default="chol")
+    p.add_argument("--device", default=None, help="cpu fallback for tests")
+    p.add_argument("--layers", default="3,4,6,3")
+    return p.parse_args()
+
+
+def main():
+    args = get_args()
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    rank = int(os.environ.get("RANK", "0"))
+    local_rank = int(os.environ.get("LOCAL_RANK", rank))
+    use_cuda = torch.cuda.is_available() and args.device != "cpu"
+    if args.device == "cpu":
+        # keep the CPU smoke path fast
+        args.batch = min(args.batch, 2)
+        args.img = min(args.img, 64)
+        args.dtype = "float32"
+
+    if world > 1:
+        dist.init_process_group(backend="nccl" if use_cuda else "gloo")
+        if use_cuda:
+            torch.cuda.set_device(local_rank)
+    device = torch.device(f"cuda:{local_rank}" if use_cuda else "cpu")
+    dtype = torch.bfloat16 if args.dtype == "bfloat16" else torch.float32
+
+    torch.manual_seed(1234 + rank)
+    from dwt_amd.models import Bottleneck, ResNetDWT
+    from dwt_amd.parallel import BucketedDataParallel
+    from dwt_amd.ops import functional as Fdwt
+
+    layers = [int(x) for x in args.layers.split(",")]
+    model = ResNetDWT(Bottleneck, layers, None, num_classes=args.num_classes,
+                      group_size=args.group_size, whiten_mode=args.whiten_mode)
+    model = model.to(device).to(dtype).train()
+
+    final_layer, rest = [], []
+    for name, param in model.named_parameters():
+        (final_layer if name.startswith("fc_out") else rest).append(param)
+    lr = 1e-2
+    optimizer = torch.optim.SGD(
+        [{"params": rest}, {"params": final_layer, "lr": lr}],
+        lr=lr * 0.1, momentum=0.9, weight_decay=5e-4)
+
+    ddp = BucketedDataParallel(model)
+
+    b = args.batch
+    n_per_gpu = 3 * b
+    data = torch.randn(n_per_gpu, 3, args.img, args.img, device=device, dtype=dtype)
+    labels = torch.randint(0, args.num_classes, (b,), device=device)
+
+    def step():
+        optimizer.zero_grad(set_to_none=True)
+        out = model(data)
+        s, t, td = torch.split(out, b, dim=0)
+        cls_loss = F.nll_loss(F.log_softmax(s.float(), dim=1), labels)
+        mec = 0.1 * Fdwt.mec_loss(t, td)
+        (cls_loss + mec).backward()
+        if ddp.enabled:
+            ddp.sync()
+        optimizer.step()
+
+    def barrier_sync():
+        if world > 1:
+            dist.barrier()
+        if use_cuda:
+            torch.cuda.synchronize()
+
+    for _ in range(args.warmup):
+        step()
+    barrier_sync()
+    t0 = time.perf_counter()
+    for _ in range(args.steps):
+        step()
+    barrier_sync()
+    elapsed = time.perf_counter() - t0
+
+    # max over ranks
+    if world > 1:
+        t = torch.tensor([elapsed], dtype=torch.float64,
+                         device=device if use_cuda else "cpu")
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        elapsed = float(t.item())
+
+    if rank == 0:
+        total_images = n_per_gpu * world * args.steps
+        value = total_images / elapsed
+        print(json.dumps({
+            "metric": "imgs_per_sec",
+            "value": value,
+            "unit": "images/sec",
+            "n_gpus": world,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": 1000.0 * elapsed / args.steps,
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": args.dtype if args.dtype != "bfloat16" else "bf16",
+            "data": "synthetic",
+            "config": {
+                "model": "resnet50-dwt-mec-3stream",
+                "global_batch": n_per_gpu * world,
+                "seq_len": args.img,
+                "parallelism": f"dp{world}",
+                "per_domain_batch": b,
+                "group_size": args.group_size,
+                "whiten_mode": args.whiten_mode,
+            },
+        }))
+
+    if world > 1:
+        dist.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()
