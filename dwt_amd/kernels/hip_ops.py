@@ -1,0 +1,322 @@
+"""GPU autograd Functions backed by the gfx950 HIP kernels.
+
+Mirrors the algorithm contract of dwt_amd/ops/functional.py exactly (these
+are numerically tested against it in tests/test_gpu_kernels.py).  Fast path
+covers group sizes {2, 4, 8} — every config the reference models use with the
+default group_size=4 — and fp32/bf16 inputs; other group sizes fall back to
+the torch implementation (warned once).
+"""
+from __future__ import annotations
+
+import warnings
+from typing import Optional, Sequence
+
+import torch
+
+from . import dispatch
+
+_FAST_G = (2, 4, 8)
+_warned = set()
+
+
+def _warn_once(key, msg):
+    if key not in _warned:
+        _warned.add(key)
+        warnings.warn(msg)
+
+
+def _ext():
+    return dispatch.ext()
+
+
+def _f32c(t):
+    return t.detach().to(torch.float32).contiguous()
+
+
+class _HipWhitenMulti(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, gamma, beta, running_means, running_vars, cfg):
+        ext = _ext()
+        parts = cfg["parts"]
+        g = x.shape[1] // cfg["num_groups"]
+        eps, momentum = cfg["eps"], cfg["momentum"]
+        training, mode, relu = cfg["training"], cfg["mode"], cfg["relu"]
+        track = cfg.get("track_running_stats", True)
+        ns_iters = cfg.get("ns_iters", 7)
+        use_batch = training or not track
+
+        x = x.contiguous()
+        n, c, h, w = x.shape
+        b = n // parts
+        n_groups = c // g
+        dev = x.device
+        m_count = b * h * w
+
+        has_affine = gamma is not None
+        gamma32 = _f32c(gamma).reshape(c) if has_affine else torch.empty(0, device=dev)
+        beta32 = _f32c(beta).reshape(c) if has_affine else torch.empty(0, device=dev)
+
+        out = torch.empty_like(x)
+        means, wmats, saved_mat = [], [], []
+        for p in range(parts):
+            xp = x[p * b:(p + 1) * b]
+            if use_batch:
+                acc = torch.zeros(n_groups * (g + g * g), device=dev, dtype=torch.float32)
+                mean = torch.empty(c, device=dev, dtype=torch.float32)
+                cov = torch.empty(n_groups, g, g, device=dev, dtype=torch.float32)
+                ext.whiten_stats(xp, acc, mean, cov, g)
+            else:
+                mean = _f32c(running_means[p]).reshape(c)
+                cov = _f32c(running_vars[p]).reshape(n_groups, g, g)
+
+            wmat = torch.empty(n_groups, g, g, device=dev, dtype=torch.float32)
+            if mode == "chol":
+                ell = torch.empty_like(wmat)
+                ext.matfn_chol_fwd(cov, wmat, ell, eps)
+                saved_mat.append(ell)
+            else:
+                ys = torch.empty(n_groups, ns_iters, g, g, device=dev, dtype=torch.float32)
+                zs = torch.empty_like(ys)
+                svals = torch.empty(n_groups, device=dev, dtype=torch.float32)
+                ext.matfn_ns_fwd(cov, wmat, ys, zs, svals, eps, ns_iters)
+                saved_mat.append((ys, zs, svals))
+
+            ext.whiten_apply(xp, mean, wmat, gamma32, beta32,
+                             out[p * b:(p + 1) * b], g, relu, has_affine)
+            means.append(mean)
+            wmats.append(wmat)
+
+            if training and track and running_means is not None:
+                with torch.no_grad():
+                    rm, rv = running_means[p], running_vars[p]
+                    rm.mul_(1.0 - momentum).add_(
+                        mean.reshape(rm.shape).to(rm.dtype), alpha=momentum)
+                    rv.mul_(1.0 - momentum).add_(
+                        cov.reshape(rv.shape).to(rv.dtype), alpha=momentum)
+
+        ctx.cfg = cfg
+        ctx.g = g
+        ctx.m_count = m_count
+        ctx.means = means
+        ctx.wmats = wmats
+        ctx.saved_mat = saved_mat
+        ctx.has_affine = has_affine
+        ctx.gamma32 = gamma32
+        ctx.save_for_backward(x, gamma, out)
+        return out
+
+    @staticmethod
+    def backward(ctx, dout):
+        ext = _ext()
+        cfg = ctx.cfg
+        parts, relu, mode, eps = cfg["parts"], cfg["relu"], cfg["mode"], cfg["eps"]
+        training = cfg["training"]
+        track = cfg.get("track_running_stats", True)
+        ns_iters = cfg.get("ns_iters", 7)
+        use_batch = training or not track
+        g = ctx.g
+        x, gamma, out = ctx.saved_tensors
+        n, c, h, w = x.shape
+        b = n // parts
+        n_groups = c // g
+        dev = x.device
+        dout = dout.contiguous()
+
+        dgb = torch.zeros(parts, 2, c, device=dev, dtype=torch.float32)
+        dx = torch.empty_like(x)
+        for p in range(parts):
+            sl = slice(p * b, (p + 1) * b)
+            xp, doutp, outp = x[sl], dout[sl], out[sl]
+            mean, wmat = ctx.means[p], ctx.wmats[p]
+            dW = torch.zeros(n_groups, g, g, device=dev, dtype=torch.float32)
+            ext.whiten_bwd_reduce(xp, doutp, outp, mean, wmat, ctx.gamma32,
+                                  dW, dgb[p].reshape(-1), g, relu, ctx.has_affine)
+            if use_batch:
+                gdb = (ctx.gamma32 * dgb[p, 1]) if ctx.has_affine else dgb[p, 1]
+                gdb = gdb.contiguous()
+                S = torch.empty(n_groups, g, g, device=dev, dtype=torch.float32)
+                corr = torch.empty(c, device=dev, dtype=torch.float32)
+                inv_m = 1.0 / ctx.m_count
+                if mode == "chol":
+                    ext.matfn_chol_bwd(dW, wmat, ctx.saved_mat[p], gdb, S, corr,
+                                       eps, inv_m)
+                else:
+                    ys, zs, svals = ctx.saved_mat[p]
+                    ext.matfn_ns_bwd(dW, wmat, ys, zs, svals, gdb, S, corr,
+                                     eps, inv_m, ns_iters)
+            else:
+                S = torch.empty(0, device=dev)
+                corr = torch.empty(0, device=dev)
+            ext.whiten_bwd_apply(xp, doutp, outp, mean, wmat, ctx.gamma32,
+                                 S, corr, dx[sl], g, relu, ctx.has_affine,
+                                 use_batch)
+
+        if ctx.has_affine:
+            dgamma = dgb[:, 0].sum(0).reshape(gamma.shape).to(gamma.dtype)
+            dbeta = dgb[:, 1].sum(0).reshape(gamma.shape).to(gamma.dtype)
+        else:
+            dgamma = dbeta = None
+        return dx, dgamma, dbeta, None, None, None
+
+
+def whiten_multi(x, gamma, beta, running_means, running_vars, cfg):
+    g = x.shape[1] // cfg["num_groups"]
+    if g not in _FAST_G or x.dtype not in (torch.float32, torch.bfloat16):
+        _warn_once(("wh", g, x.dtype),
+                   f"dwt_amd: whiten_multi g={g} dtype={x.dtype} uses the "
+                   "torch path (HIP fast path covers g in {2,4,8}, fp32/bf16)")
+        from ..ops.functional import WhitenMulti
+        return WhitenMulti.apply(x, gamma, beta, running_means, running_vars, cfg)
+    return _HipWhitenMulti.apply(x, gamma, beta, running_means, running_vars, cfg)
+
+
+class _HipBatchNormMulti(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, gamma, beta, running_means, running_vars, cfg):
+        ext = _ext()
+        parts = cfg["parts"]
+        eps, momentum = cfg["eps"], cfg["momentum"]
+        training, relu = cfg["training"], cfg["relu"]
+        track = cfg.get("track_running_stats", True)
+        use_batch = training or not track
+
+        spatial = x.dim() == 4
+        x = x.contiguous()
+        n, c = x.shape[0], x.shape[1]
+        b = n // parts
+        dev = x.device
+        cnt = (x.numel() // parts) // c
+
+        has_affine = gamma is not None
+        gamma32 = _f32c(gamma).reshape(c) if has_affine else torch.empty(0, device=dev)
+        beta32 = _f32c(beta).reshape(c) if has_affine else torch.empty(0, device=dev)
+
+        out = torch.empty_like(x)
+        means, istds = [], []
+        for p in range(parts):
+            xp = x[p * b:(p + 1) * b]
+            if use_batch:
+                acc = torch.zeros(2 * c, device=dev, dtype=torch.float32)
+                mean = torch.empty(c, device=dev, dtype=torch.float32)
+                istd = torch.empty(c, device=dev, dtype=torch.float32)
+                var_unb = torch.empty(c, device=dev, dtype=torch.float32)
+                ext.bn_stats(xp, acc, mean, istd, var_unb, eps)
+                if training and track and running_means is not None:
+                    with torch.no_grad():
+                        rm, rv = running_means[p], running_vars[p]
+                        rm.mul_(1 - momentum).add_(mean.to(rm.dtype), alpha=momentum)
+                        rv.mul_(1 - momentum).add_(var_unb.to(rv.dtype), alpha=momentum)
+            else:
+                mean = _f32c(running_means[p]).reshape(c)
+                var = _f32c(running_vars[p]).reshape(c)
+                istd = torch.rsqrt(var + eps)
+            ext.bn_apply(xp, mean, istd, gamma32, beta32,
+                         out[p * b:(p + 1) * b], relu, has_affine)
+            means.append(mean)
+            istds.append(istd)
+
+        ctx.cfg = cfg
+        ctx.spatial = spatial
+        ctx.means = means
+        ctx.istds = istds
+        ctx.has_affine = has_affine
+        ctx.gamma32 = gamma32
+        ctx.use_batch = use_batch
+        ctx.save_for_backward(x, gamma, out)
+        return out
+
+    @staticmethod
+    def backward(ctx, dout):
+        ext = _ext()
+        cfg = ctx.cfg
+        parts, relu = cfg["parts"], cfg["relu"]
+        x, gamma, out = ctx.saved_tensors
+        n, c = x.shape[0], x.shape[1]
+        b = n // parts
+        dev = x.device
+        dout = dout.contiguous()
+
+        sums = torch.zeros(parts, 2, c, device=dev, dtype=torch.float32)
+        dx = torch.empty_like(x)
+        for p in range(parts):
+            sl = slice(p * b, (p + 1) * b)
+            xp, doutp, outp = x[sl], dout[sl], out[sl]
+            ext.bn_bwd_reduce(xp, doutp, outp, ctx.means[p], ctx.istds[p],
+                              sums[p].reshape(-1), relu)
+            ext.bn_bwd_apply(xp, doutp, outp, ctx.means[p], ctx.istds[p],
+                             ctx.gamma32, sums[p].reshape(-1), dx[sl], relu,
+                             ctx.has_affine, ctx.use_batch)
+
+        if ctx.has_affine:
+            # dgamma = sum dy * xhat ; dbeta = sum dy  (sums[:,1] is dy*xhat)
+            dgamma = sums[:, 1].sum(0).reshape(gamma.shape).to(gamma.dtype)
+            dbeta = sums[:, 0].sum(0).reshape(gamma.shape).to(gamma.dtype)
+        else:
+            dgamma = dbeta = None
+        return dx, dgamma, dbeta, None, None, None
+
+
+def batch_norm_multi(x, gamma, beta, running_means, running_vars, cfg):
+    if x.dtype not in (torch.float32, torch.bfloat16):
+        from ..ops.functional import BatchNormMulti
+        return BatchNormMulti.apply(x, gamma, beta, running_means, running_vars, cfg)
+    return _HipBatchNormMulti.apply(x, gamma, beta, running_means, running_vars, cfg)
+
+
+class _HipMecLoss(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, y):
+        ext = _ext()
+        x32 = x.detach().to(torch.float32).contiguous()
+        y32 = y.detach().to(torch.float32).contiguous()
+        n, k = x32.shape
+        lx = torch.empty_like(x32)
+        ly = torch.empty_like(y32)
+        amin = torch.empty(n, device=x.device, dtype=torch.int32)
+        loss = torch.zeros(1, device=x.device, dtype=torch.float32)
+        ext.mec_fwd(x32, y32, lx, ly, amin, loss)
+        ctx.save_for_backward(lx, ly, amin)
+        ctx.dtypes = (x.dtype, y.dtype)
+        return loss[0]
+
+    @staticmethod
+    def backward(ctx, dloss):
+        ext = _ext()
+        lx, ly, amin = ctx.saved_tensors
+        gscale = dloss.detach().to(torch.float32).reshape(1).contiguous()
+        dx = torch.empty_like(lx)
+        dy = torch.empty_like(ly)
+        ext.mec_bwd(lx, ly, amin, gscale, dx, dy)
+        return dx.to(ctx.dtypes[0]), dy.to(ctx.dtypes[1])
+
+
+class _HipEntropyLoss(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x):
+        ext = _ext()
+        x32 = x.detach().to(torch.float32).contiguous()
+        n, k = x32.shape
+        q = torch.empty_like(x32)
+        hper = torch.empty(n, device=x.device, dtype=torch.float32)
+        loss = torch.zeros(1, device=x.device, dtype=torch.float32)
+        ext.entropy_fwd(x32, q, hper, loss)
+        ctx.save_for_backward(q, hper)
+        ctx.dtype = x.dtype
+        return loss[0]
+
+    @staticmethod
+    def backward(ctx, dloss):
+        ext = _ext()
+        q, hper = ctx.saved_tensors
+        gscale = dloss.detach().to(torch.float32).reshape(1).contiguous()
+        dx = torch.empty_like(q)
+        ext.entropy_bwd(q, hper, gscale, dx)
+        return dx.to(ctx.dtype)
+
+
+def mec_loss(x, y):
+    return _HipMecLoss.apply(x, y)
+
+
+def entropy_loss(x):
+    return _HipEntropyLoss.apply(x)

@@ -43,6 +43,15 @@ class _DomainBatchNorm(nn.Module):
             self.register_parameter("num_batches_tracked", None)
         self.reset_parameters()
 
+    def _apply(self, fn, recurse=True):
+        # statistics buffers stay fp32 under model-wide bf16 conversion
+        super()._apply(fn, recurse)
+        for name in ("running_mean", "running_var"):
+            b = self._buffers.get(name)
+            if b is not None and b.is_floating_point() and b.dtype != torch.float32:
+                self._buffers[name] = b.float()
+        return self
+
     def reset_running_stats(self):
         if self.track_running_stats:
             self.num_batches_tracked.zero_()

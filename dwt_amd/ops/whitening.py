@@ -43,6 +43,16 @@ class _Whitening(nn.Module):
             self.register_buffer("running_variance",
                                  torch.ones(self.num_groups, self.group_size, self.group_size))
 
+    def _apply(self, fn, recurse=True):
+        # statistics buffers stay fp32 even when the model runs bf16 —
+        # a bf16 EMA of covariances is too coarse
+        super()._apply(fn, recurse)
+        for name in ("running_mean", "running_variance"):
+            b = self._buffers.get(name)
+            if b is not None and b.is_floating_point() and b.dtype != torch.float32:
+                self._buffers[name] = b.float()
+        return self
+
     def _check_input_dim(self, x):
         raise NotImplementedError
 
