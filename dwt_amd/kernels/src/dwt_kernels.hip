@@ -110,10 +110,11 @@ DEV_INLINE float wave_reduce_sum(float v) {
 // acc layout per group: [g sums][g*g product sums]
 // ===========================================================================
 
-template <typename T, int G>
+template <typename T, int G, bool VECTOR>
 __global__ void whiten_stats_partial_kernel(
     const T* __restrict__ x, float* __restrict__ acc,
     int B, int C, int64_t HW, int64_t M) {
+  constexpr int VW = VECTOR ? VecTraits<T>::W : 1;
   const int grp = blockIdx.y;
   const int c0 = grp * G;
   float s[G];
@@ -124,18 +125,26 @@ __global__ void whiten_stats_partial_kernel(
 #pragma unroll
     for (int j = 0; j < G; ++j) p[i][j] = 0.f;
   }
+  const int64_t nvec = M / VW;
   const int64_t stride = (int64_t)gridDim.x * blockDim.x;
-  for (int64_t m = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; m < M; m += stride) {
-    const int64_t n = m / HW, hw = m - n * HW;
+  for (int64_t mv = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; mv < nvec; mv += stride) {
+    const int64_t el = mv * VW;
+    const int64_t n = el / HW, hw = el - n * HW;
     const T* base = x + ((n * C + c0) * HW + hw);
-    float v[G];
+    float v[G][VW];
 #pragma unroll
-    for (int j = 0; j < G; ++j) v[j] = ldf(base + j * HW);
+    for (int j = 0; j < G; ++j) {
+      if (VECTOR) VecTraits<T>::load(base + (int64_t)j * HW, v[j]);
+      else v[j][0] = ldf(base + (int64_t)j * HW);
+    }
 #pragma unroll
-    for (int i = 0; i < G; ++i) {
-      s[i] += v[i];
+    for (int k = 0; k < VW; ++k) {
 #pragma unroll
-      for (int j = 0; j <= i; ++j) p[i][j] += v[i] * v[j];
+      for (int i = 0; i < G; ++i) {
+        s[i] += v[i][k];
+#pragma unroll
+        for (int j = 0; j <= i; ++j) p[i][j] += v[i][k] * v[j][k];
+      }
     }
   }
   float* gacc = acc + (int64_t)grp * (G + G * G);
@@ -974,9 +983,17 @@ inline int64_t elementwise_blocks(int64_t work, int threads) {
   return std::max<int64_t>(1, (work + threads - 1) / threads);
 }
 
-inline int64_t reduce_blocks(int64_t work_per_group, int threads, int vec) {
-  int64_t b = (work_per_group + (int64_t)threads * vec - 1) / ((int64_t)threads * vec);
-  return std::min<int64_t>(std::max<int64_t>(b, 1), 2048);
+// Blocks along the reduction axis for a (groups x work) reduction grid.
+// Sized so the TOTAL grid is ~4096 workgroups (fills 256 CUs several times
+// over) — per-group caps must shrink as the group count grows, or the
+// all-groups grid explodes and the final atomics dominate (measured: the
+// uncapped version put whiten_stats at 46% of step time).
+inline int64_t reduce_blocks(int64_t work_per_group, int threads, int vec,
+                             int64_t n_groups) {
+  const int64_t want = (work_per_group + (int64_t)threads * vec - 1) /
+                       ((int64_t)threads * vec);
+  const int64_t cap = std::max<int64_t>(1, 4096 / std::max<int64_t>(n_groups, 1));
+  return std::min<int64_t>(std::max<int64_t>(want, 1), cap);
 }
 
 // dispatch over float + bf16 only (no fp64 kernels on the GPU path)
@@ -1011,18 +1028,25 @@ void whiten_stats(Tensor x, Tensor acc, Tensor mean, Tensor cov, int64_t g) {
   const int64_t M = (int64_t)B * HW;
   const int n_groups = C / g;
   DISPATCH_FT(x, "whiten_stats", [&] {
+    constexpr int VW = dwt::VecTraits<scalar_t>::W;
+    const bool vec = can_vectorize<scalar_t>(x, HW) && (M % VW == 0);
     const int threads = 256;
-    dim3 grid(reduce_blocks(M, threads, 1), n_groups);
-    auto launch = [&](auto gconst) {
+    dim3 grid(reduce_blocks(M, threads, vec ? VW : 1, n_groups), n_groups);
+    auto launch = [&](auto gconst, auto vconst) {
       constexpr int G = decltype(gconst)::value;
-      hipLaunchKernelGGL((dwt::whiten_stats_partial_kernel<scalar_t, G>), grid,
+      constexpr bool V = decltype(vconst)::value;
+      hipLaunchKernelGGL((dwt::whiten_stats_partial_kernel<scalar_t, G, V>), grid,
                          dim3(threads), 0, cur_stream(),
                          x.data_ptr<scalar_t>(), acc.data_ptr<float>(), B, C, HW, M);
     };
+    auto pick_v = [&](auto gconst) {
+      if (vec) launch(gconst, std::true_type{});
+      else launch(gconst, std::false_type{});
+    };
     switch (g) {
-      case 2: launch(std::integral_constant<int, 2>{}); break;
-      case 4: launch(std::integral_constant<int, 4>{}); break;
-      case 8: launch(std::integral_constant<int, 8>{}); break;
+      case 2: pick_v(std::integral_constant<int, 2>{}); break;
+      case 4: pick_v(std::integral_constant<int, 4>{}); break;
+      case 8: pick_v(std::integral_constant<int, 8>{}); break;
       default: TORCH_CHECK(false, "whiten_stats: unsupported group size ", g);
     }
   });
@@ -1128,7 +1152,7 @@ void whiten_bwd_reduce(Tensor x, Tensor dout, Tensor out, Tensor mean, Tensor W,
     auto launch = [&](auto gconst, auto vconst) {
       constexpr int G = decltype(gconst)::value;
       constexpr bool V = decltype(vconst)::value;
-      dim3 grid(reduce_blocks(M, threads, V ? VW : 1), n_groups);
+      dim3 grid(reduce_blocks(M, threads, V ? VW : 1, n_groups), n_groups);
       hipLaunchKernelGGL((dwt::whiten_bwd_reduce_kernel<scalar_t, G, V>), grid,
                          dim3(threads), 0, cur_stream(), x.data_ptr<scalar_t>(),
                          dout.data_ptr<scalar_t>(), out.data_ptr<scalar_t>(),
@@ -1200,7 +1224,7 @@ void bn_stats(Tensor x, Tensor acc, Tensor mean, Tensor istd, Tensor var_unb,
     constexpr int VW = dwt::VecTraits<scalar_t>::W;
     const bool vec = can_vectorize<scalar_t>(x, HW) && (M % VW == 0);
     const int threads = 256;
-    dim3 grid(reduce_blocks(M, threads, vec ? VW : 1), C);
+    dim3 grid(reduce_blocks(M, threads, vec ? VW : 1, C), C);
     if (vec)
       hipLaunchKernelGGL((dwt::bn_stats_partial_kernel<scalar_t, true>), grid,
                          dim3(threads), 0, cur_stream(), x.data_ptr<scalar_t>(),
@@ -1253,7 +1277,7 @@ void bn_bwd_reduce(Tensor x, Tensor dout, Tensor out, Tensor mean, Tensor istd,
     const bool vec = can_vectorize<scalar_t>(x, HW) &&
                      can_vectorize<scalar_t>(dout, HW) && (M % VW == 0);
     const int threads = 256;
-    dim3 grid(reduce_blocks(M, threads, vec ? VW : 1), C);
+    dim3 grid(reduce_blocks(M, threads, vec ? VW : 1, C), C);
     auto lp = [&](auto vconst) {
       constexpr bool V = decltype(vconst)::value;
       hipLaunchKernelGGL((dwt::bn_bwd_reduce_kernel<scalar_t, V>), grid,
