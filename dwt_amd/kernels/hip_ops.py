@@ -53,8 +53,8 @@ class _HipWhitenMulti(torch.autograd.Function):
         m_count = b * h * w
 
         has_affine = gamma is not None
-        gamma32 = _f32c(gamma).reshape(c) if has_affine else torch.empty(0, device=dev)
-        beta32 = _f32c(beta).reshape(c) if has_affine else torch.empty(0, device=dev)
+        gflat = gamma.detach().reshape(c).contiguous() if has_affine else torch.empty(0, device=dev, dtype=x.dtype)
+        bflat = beta.detach().reshape(c).contiguous() if has_affine else torch.empty(0, device=dev, dtype=x.dtype)
 
         out = torch.empty_like(x)
         means, wmats, saved_mat = [], [], []
@@ -81,7 +81,7 @@ class _HipWhitenMulti(torch.autograd.Function):
                 ext.matfn_ns_fwd(cov, wmat, ys, zs, svals, eps, ns_iters)
                 saved_mat.append((ys, zs, svals))
 
-            ext.whiten_apply(xp, mean, wmat, gamma32, beta32,
+            ext.whiten_apply(xp, mean, wmat, gflat, bflat,
                              out[p * b:(p + 1) * b], g, relu, has_affine)
             means.append(mean)
             wmats.append(wmat)
@@ -89,10 +89,14 @@ class _HipWhitenMulti(torch.autograd.Function):
             if training and track and running_means is not None:
                 with torch.no_grad():
                     rm, rv = running_means[p], running_vars[p]
-                    rm.mul_(1.0 - momentum).add_(
-                        mean.reshape(rm.shape).to(rm.dtype), alpha=momentum)
-                    rv.mul_(1.0 - momentum).add_(
-                        cov.reshape(rv.shape).to(rv.dtype), alpha=momentum)
+                    if rm.dtype == torch.float32 and rm.is_contiguous() \
+                            and rv.dtype == torch.float32 and rv.is_contiguous():
+                        ext.ema_update(rm, mean, rv, cov, momentum)
+                    else:
+                        rm.mul_(1.0 - momentum).add_(
+                            mean.reshape(rm.shape).to(rm.dtype), alpha=momentum)
+                        rv.mul_(1.0 - momentum).add_(
+                            cov.reshape(rv.shape).to(rv.dtype), alpha=momentum)
 
         ctx.cfg = cfg
         ctx.g = g
@@ -101,7 +105,7 @@ class _HipWhitenMulti(torch.autograd.Function):
         ctx.wmats = wmats
         ctx.saved_mat = saved_mat
         ctx.has_affine = has_affine
-        ctx.gamma32 = gamma32
+        ctx.gflat = gflat
         ctx.save_for_backward(x, gamma, out)
         return out
 
@@ -129,10 +133,10 @@ class _HipWhitenMulti(torch.autograd.Function):
             xp, doutp, outp = x[sl], dout[sl], out[sl]
             mean, wmat = ctx.means[p], ctx.wmats[p]
             dW = torch.zeros(n_groups, g, g, device=dev, dtype=torch.float32)
-            ext.whiten_bwd_reduce(xp, doutp, outp, mean, wmat, ctx.gamma32,
+            ext.whiten_bwd_reduce(xp, doutp, outp, mean, wmat, ctx.gflat,
                                   dW, dgb[p].reshape(-1), g, relu, ctx.has_affine)
             if use_batch:
-                gdb = (ctx.gamma32 * dgb[p, 1]) if ctx.has_affine else dgb[p, 1]
+                gdb = (ctx.gflat.float() * dgb[p, 1]) if ctx.has_affine else dgb[p, 1]
                 gdb = gdb.contiguous()
                 S = torch.empty(n_groups, g, g, device=dev, dtype=torch.float32)
                 corr = torch.empty(c, device=dev, dtype=torch.float32)
@@ -147,7 +151,7 @@ class _HipWhitenMulti(torch.autograd.Function):
             else:
                 S = torch.empty(0, device=dev)
                 corr = torch.empty(0, device=dev)
-            ext.whiten_bwd_apply(xp, doutp, outp, mean, wmat, ctx.gamma32,
+            ext.whiten_bwd_apply(xp, doutp, outp, mean, wmat, ctx.gflat,
                                  S, corr, dx[sl], g, relu, ctx.has_affine,
                                  use_batch)
 
@@ -188,8 +192,8 @@ class _HipBatchNormMulti(torch.autograd.Function):
         cnt = (x.numel() // parts) // c
 
         has_affine = gamma is not None
-        gamma32 = _f32c(gamma).reshape(c) if has_affine else torch.empty(0, device=dev)
-        beta32 = _f32c(beta).reshape(c) if has_affine else torch.empty(0, device=dev)
+        gflat = gamma.detach().reshape(c).contiguous() if has_affine else torch.empty(0, device=dev, dtype=x.dtype)
+        bflat = beta.detach().reshape(c).contiguous() if has_affine else torch.empty(0, device=dev, dtype=x.dtype)
 
         out = torch.empty_like(x)
         means, istds = [], []
@@ -204,13 +208,17 @@ class _HipBatchNormMulti(torch.autograd.Function):
                 if training and track and running_means is not None:
                     with torch.no_grad():
                         rm, rv = running_means[p], running_vars[p]
-                        rm.mul_(1 - momentum).add_(mean.to(rm.dtype), alpha=momentum)
-                        rv.mul_(1 - momentum).add_(var_unb.to(rv.dtype), alpha=momentum)
+                        if rm.dtype == torch.float32 and rm.is_contiguous() \
+                                and rv.dtype == torch.float32 and rv.is_contiguous():
+                            ext.ema_update(rm, mean, rv, var_unb, momentum)
+                        else:
+                            rm.mul_(1 - momentum).add_(mean.to(rm.dtype), alpha=momentum)
+                            rv.mul_(1 - momentum).add_(var_unb.to(rv.dtype), alpha=momentum)
             else:
                 mean = _f32c(running_means[p]).reshape(c)
                 var = _f32c(running_vars[p]).reshape(c)
                 istd = torch.rsqrt(var + eps)
-            ext.bn_apply(xp, mean, istd, gamma32, beta32,
+            ext.bn_apply(xp, mean, istd, gflat, bflat,
                          out[p * b:(p + 1) * b], relu, has_affine)
             means.append(mean)
             istds.append(istd)
@@ -220,7 +228,7 @@ class _HipBatchNormMulti(torch.autograd.Function):
         ctx.means = means
         ctx.istds = istds
         ctx.has_affine = has_affine
-        ctx.gamma32 = gamma32
+        ctx.gflat = gflat
         ctx.use_batch = use_batch
         ctx.save_for_backward(x, gamma, out)
         return out
@@ -244,7 +252,7 @@ class _HipBatchNormMulti(torch.autograd.Function):
             ext.bn_bwd_reduce(xp, doutp, outp, ctx.means[p], ctx.istds[p],
                               sums[p].reshape(-1), relu)
             ext.bn_bwd_apply(xp, doutp, outp, ctx.means[p], ctx.istds[p],
-                             ctx.gamma32, sums[p].reshape(-1), dx[sl], relu,
+                             ctx.gflat, sums[p].reshape(-1), dx[sl], relu,
                              ctx.has_affine, ctx.use_batch)
 
         if ctx.has_affine:

@@ -105,6 +105,32 @@ DEV_INLINE float wave_reduce_sum(float v) {
   return v;
 }
 
+// Block-level reduction of NV per-thread partials into global accumulators:
+// wave shuffle -> LDS (one row per wave) -> one atomicAdd per value per
+// BLOCK (not per wave) — atomics on the shared accumulator addresses were
+// the measured tail of every reduce kernel (Guideline 12).
+// lds must hold (blockDim.x/64) * NV floats; idx(k) maps partial k to its
+// global accumulator offset.
+template <int NV, typename IdxFn>
+DEV_INLINE void block_reduce_atomic(float (&vals)[NV], float* lds,
+                                    float* gout, IdxFn idx) {
+  const int lane = threadIdx.x & 63;
+  const int wid = threadIdx.x >> 6;
+  const int nwaves = blockDim.x >> 6;
+#pragma unroll
+  for (int k = 0; k < NV; ++k) {
+    const float r = wave_reduce_sum(vals[k]);
+    if (lane == 0) lds[wid * NV + k] = r;
+  }
+  __syncthreads();
+  for (int k = threadIdx.x; k < NV; k += blockDim.x) {
+    float acc = 0.f;
+    for (int w = 0; w < nwaves; ++w) acc += lds[w * NV + k];
+    atomicAdd(&gout[idx(k)], acc);
+  }
+  __syncthreads();  // lds reusable by caller afterwards
+}
+
 // ===========================================================================
 // Whitening: fused mean + covariance partial pass (register-blocked, g <= 8)
 // acc layout per group: [g sums][g*g product sums]
@@ -148,17 +174,21 @@ __global__ void whiten_stats_partial_kernel(
     }
   }
   float* gacc = acc + (int64_t)grp * (G + G * G);
-  const int lane = threadIdx.x & 63;
+  constexpr int NTRI = G * (G + 1) / 2;
+  __shared__ float red_lds[4 * (G + NTRI)];
+  float vals[G + NTRI];
 #pragma unroll
-  for (int i = 0; i < G; ++i) {
-    float r = wave_reduce_sum(s[i]);
-    if (lane == 0) atomicAdd(&gacc[i], r);
+  for (int i = 0; i < G; ++i) vals[i] = s[i];
 #pragma unroll
-    for (int j = 0; j <= i; ++j) {
-      float q = wave_reduce_sum(p[i][j]);
-      if (lane == 0) atomicAdd(&gacc[G + i * G + j], q);
-    }
-  }
+  for (int i = 0; i < G; ++i)
+#pragma unroll
+    for (int j = 0; j <= i; ++j) vals[G + i * (i + 1) / 2 + j] = p[i][j];
+  block_reduce_atomic<G + NTRI>(vals, red_lds, gacc, [](int k) {
+    if (k < G) return k;
+    int kk = k - G, i = 0;
+    while (kk > i) { kk -= (i + 1); ++i; }
+    return G + i * G + kk;
+  });
 }
 
 // finalize: mean_c, cov[g,i,j] = E[x_i x_j] - mu_i mu_j  (full symmetric)
@@ -469,8 +499,8 @@ __global__ void matfn_ns_bwd_kernel(
 template <typename T, int G, bool VECTOR>
 __global__ void whiten_apply_kernel(
     const T* __restrict__ x, const float* __restrict__ mean,
-    const float* __restrict__ W, const float* __restrict__ gamma,
-    const float* __restrict__ beta, T* __restrict__ out,
+    const float* __restrict__ W, const T* __restrict__ gamma,
+    const T* __restrict__ beta, T* __restrict__ out,
     int C, int64_t HW, int relu, int has_affine) {
   constexpr int VW = VECTOR ? VecTraits<T>::W : 1;
   const int grp = blockIdx.z;
@@ -483,8 +513,8 @@ __global__ void whiten_apply_kernel(
 #pragma unroll
   for (int i = 0; i < G; ++i) {
     m[i] = mean[c0 + i];
-    gm[i] = has_affine ? gamma[c0 + i] : 1.f;
-    bt[i] = has_affine ? beta[c0 + i] : 0.f;
+    gm[i] = has_affine ? ldf(gamma + c0 + i) : 1.f;
+    bt[i] = has_affine ? ldf(beta + c0 + i) : 0.f;
 #pragma unroll
     for (int j = 0; j < G; ++j) Wr[i][j] = W[((int64_t)grp * G + i) * G + j];
   }
@@ -532,7 +562,7 @@ template <typename T, int G, bool VECTOR>
 __global__ void whiten_bwd_reduce_kernel(
     const T* __restrict__ x, const T* __restrict__ dout,
     const T* __restrict__ out, const float* __restrict__ mean,
-    const float* __restrict__ W, const float* __restrict__ gamma,
+    const float* __restrict__ W, const T* __restrict__ gamma,
     float* __restrict__ dWacc, float* __restrict__ dgb,
     int B, int C, int64_t HW, int64_t M, int relu, int has_affine) {
   constexpr int VW = VECTOR ? VecTraits<T>::W : 1;
@@ -542,7 +572,7 @@ __global__ void whiten_bwd_reduce_kernel(
 #pragma unroll
   for (int i = 0; i < G; ++i) {
     m[i] = mean[c0 + i];
-    gm[i] = has_affine ? gamma[c0 + i] : 1.f;
+    gm[i] = has_affine ? ldf(gamma + c0 + i) : 1.f;
 #pragma unroll
     for (int j = 0; j < G; ++j) Wr[i][j] = W[((int64_t)grp * G + i) * G + j];
   }
@@ -598,19 +628,23 @@ __global__ void whiten_bwd_reduce_kernel(
     }
   }
 
-  const int lane = threadIdx.x & 63;
   float* gdW = dWacc + (int64_t)grp * G * G;
+  __shared__ float red_lds[4 * G * G];
+  {
+    float vals[2 * G];
 #pragma unroll
-  for (int i = 0; i < G; ++i) {
-    float r = wave_reduce_sum(dg[i]);
-    if (lane == 0) atomicAdd(&dgb[c0 + i], r);
-    r = wave_reduce_sum(db[i]);
-    if (lane == 0) atomicAdd(&dgb[C + c0 + i], r);
+    for (int i = 0; i < G; ++i) { vals[i] = dg[i]; vals[G + i] = db[i]; }
+    block_reduce_atomic<2 * G>(vals, red_lds, dgb, [c0, C](int k) {
+      return k < G ? c0 + k : C + c0 + (k - G);
+    });
+  }
+  {
+    float vals[G * G];
 #pragma unroll
-    for (int j = 0; j < G; ++j) {
-      r = wave_reduce_sum(dWl[i][j]);
-      if (lane == 0) atomicAdd(&gdW[i * G + j], r);
-    }
+    for (int i = 0; i < G; ++i)
+#pragma unroll
+      for (int j = 0; j < G; ++j) vals[i * G + j] = dWl[i][j];
+    block_reduce_atomic<G * G>(vals, red_lds, gdW, [](int k) { return k; });
   }
 }
 
@@ -623,7 +657,7 @@ template <typename T, int G, bool VECTOR>
 __global__ void whiten_bwd_apply_kernel(
     const T* __restrict__ x, const T* __restrict__ dout,
     const T* __restrict__ out, const float* __restrict__ mean,
-    const float* __restrict__ W, const float* __restrict__ gamma,
+    const float* __restrict__ W, const T* __restrict__ gamma,
     const float* __restrict__ S, const float* __restrict__ corr,
     T* __restrict__ dx, int C, int64_t HW, int relu, int has_affine,
     int train_stats) {
@@ -638,7 +672,7 @@ __global__ void whiten_bwd_apply_kernel(
 #pragma unroll
   for (int i = 0; i < G; ++i) {
     m[i] = mean[c0 + i];
-    gm[i] = has_affine ? gamma[c0 + i] : 1.f;
+    gm[i] = has_affine ? ldf(gamma + c0 + i) : 1.f;
     cr[i] = train_stats ? corr[c0 + i] : 0.f;
 #pragma unroll
     for (int j = 0; j < G; ++j) {
@@ -717,12 +751,11 @@ __global__ void bn_stats_partial_kernel(
       s += v; ss += v * v;
     }
   }
-  s = wave_reduce_sum(s);
-  ss = wave_reduce_sum(ss);
-  if ((threadIdx.x & 63) == 0) {
-    atomicAdd(&acc[c], s);
-    atomicAdd(&acc[C + c], ss);
-  }
+  __shared__ float red_lds[4 * 2];
+  float vals[2] = {s, ss};
+  block_reduce_atomic<2>(vals, red_lds, acc, [c, C](int k) {
+    return k == 0 ? c : C + c;
+  });
 }
 
 __global__ void bn_stats_final_kernel(
@@ -741,8 +774,8 @@ __global__ void bn_stats_final_kernel(
 template <typename T, bool VECTOR>
 __global__ void bn_apply_kernel(
     const T* __restrict__ x, const float* __restrict__ mean,
-    const float* __restrict__ istd, const float* __restrict__ gamma,
-    const float* __restrict__ beta, T* __restrict__ out,
+    const float* __restrict__ istd, const T* __restrict__ gamma,
+    const T* __restrict__ beta, T* __restrict__ out,
     int C, int64_t HW, int relu, int has_affine) {
   constexpr int VW = VECTOR ? VecTraits<T>::W : 1;
   const int c = blockIdx.z;
@@ -750,8 +783,8 @@ __global__ void bn_apply_kernel(
   const int64_t hw0 = ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) * VW;
   if (hw0 >= HW) return;
   const float mu = mean[c], is = istd[c];
-  const float gm = has_affine ? gamma[c] : 1.f;
-  const float bt = has_affine ? beta[c] : 0.f;
+  const float gm = has_affine ? ldf(gamma + c) : 1.f;
+  const float bt = has_affine ? ldf(beta + c) : 0.f;
   const int64_t off = ((int64_t)n * C + c) * HW + hw0;
   float v[VW];
   if (VECTOR) VecTraits<T>::load(x + off, v);
@@ -802,19 +835,18 @@ __global__ void bn_bwd_reduce_kernel(
       s_dyxh += dv[k] * (xv[k] - mu) * is;
     }
   }
-  s_dy = wave_reduce_sum(s_dy);
-  s_dyxh = wave_reduce_sum(s_dyxh);
-  if ((threadIdx.x & 63) == 0) {
-    atomicAdd(&sums[c], s_dy);
-    atomicAdd(&sums[C + c], s_dyxh);
-  }
+  __shared__ float red_lds[4 * 2];
+  float vals[2] = {s_dy, s_dyxh};
+  block_reduce_atomic<2>(vals, red_lds, sums, [c, C](int k) {
+    return k == 0 ? c : C + c;
+  });
 }
 
 template <typename T, bool VECTOR>
 __global__ void bn_bwd_apply_kernel(
     const T* __restrict__ x, const T* __restrict__ dout,
     const T* __restrict__ out, const float* __restrict__ mean,
-    const float* __restrict__ istd, const float* __restrict__ gamma,
+    const float* __restrict__ istd, const T* __restrict__ gamma,
     const float* __restrict__ sums, T* __restrict__ dx,
     int C, int64_t HW, float inv_m, int relu, int has_affine, int use_batch) {
   constexpr int VW = VECTOR ? VecTraits<T>::W : 1;
@@ -823,7 +855,7 @@ __global__ void bn_bwd_apply_kernel(
   const int64_t hw0 = ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) * VW;
   if (hw0 >= HW) return;
   const float mu = mean[c], is = istd[c];
-  const float gm = has_affine ? gamma[c] : 1.f;
+  const float gm = has_affine ? ldf(gamma + c) : 1.f;
   const float m_dy = sums[c] * inv_m;
   const float m_dyxh = sums[C + c] * inv_m;
   const int64_t off = ((int64_t)n * C + c) * HW + hw0;
@@ -856,6 +888,25 @@ __global__ void bn_bwd_apply_kernel(
   }
   if (VECTOR) VecTraits<T>::store(dx + off, dv);
   else stf(dx + off, dv[0]);
+}
+
+// ===========================================================================
+// Fused EMA update of two running buffers in one launch:
+//   rm = (1-m) rm + m * a   (n1 elems),  rv = (1-m) rv + m * b  (n2 elems)
+// ===========================================================================
+
+__global__ void ema_update_kernel(float* __restrict__ rm,
+                                  const float* __restrict__ a, int64_t n1,
+                                  float* __restrict__ rv,
+                                  const float* __restrict__ b, int64_t n2,
+                                  float momentum) {
+  const int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  if (i < n1) {
+    rm[i] = (1.f - momentum) * rm[i] + momentum * a[i];
+  } else if (i < n1 + n2) {
+    const int64_t j = i - n1;
+    rv[j] = (1.f - momentum) * rv[j] + momentum * b[j];
+  }
 }
 
 // ===========================================================================
@@ -1119,8 +1170,8 @@ void whiten_apply(Tensor x, Tensor mean, Tensor W, Tensor gamma, Tensor beta,
       hipLaunchKernelGGL((dwt::whiten_apply_kernel<scalar_t, G, V>), grid,
                          dim3(threads), 0, cur_stream(), x.data_ptr<scalar_t>(),
                          mean.data_ptr<float>(), W.data_ptr<float>(),
-                         has_affine ? gamma.data_ptr<float>() : nullptr,
-                         has_affine ? beta.data_ptr<float>() : nullptr,
+                         has_affine ? gamma.data_ptr<scalar_t>() : nullptr,
+                         has_affine ? beta.data_ptr<scalar_t>() : nullptr,
                          out.data_ptr<scalar_t>(), C, HW, relu ? 1 : 0,
                          has_affine ? 1 : 0);
     };
@@ -1157,7 +1208,7 @@ void whiten_bwd_reduce(Tensor x, Tensor dout, Tensor out, Tensor mean, Tensor W,
                          dim3(threads), 0, cur_stream(), x.data_ptr<scalar_t>(),
                          dout.data_ptr<scalar_t>(), out.data_ptr<scalar_t>(),
                          mean.data_ptr<float>(), W.data_ptr<float>(),
-                         has_affine ? gamma.data_ptr<float>() : nullptr,
+                         has_affine ? gamma.data_ptr<scalar_t>() : nullptr,
                          dWacc.data_ptr<float>(), dgb.data_ptr<float>(), B, C,
                          HW, M, relu ? 1 : 0, has_affine ? 1 : 0);
     };
@@ -1195,7 +1246,7 @@ void whiten_bwd_apply(Tensor x, Tensor dout, Tensor out, Tensor mean, Tensor W,
                          dim3(threads), 0, cur_stream(), x.data_ptr<scalar_t>(),
                          dout.data_ptr<scalar_t>(), out.data_ptr<scalar_t>(),
                          mean.data_ptr<float>(), W.data_ptr<float>(),
-                         has_affine ? gamma.data_ptr<float>() : nullptr,
+                         has_affine ? gamma.data_ptr<scalar_t>() : nullptr,
                          S.data_ptr<float>(), corr.data_ptr<float>(),
                          dx.data_ptr<scalar_t>(), C, HW, relu ? 1 : 0,
                          has_affine ? 1 : 0, train_stats ? 1 : 0);
@@ -1258,8 +1309,8 @@ void bn_apply(Tensor x, Tensor mean, Tensor istd, Tensor gamma, Tensor beta,
       hipLaunchKernelGGL((dwt::bn_apply_kernel<scalar_t, V>), grid, dim3(threads),
                          0, cur_stream(), x.data_ptr<scalar_t>(),
                          mean.data_ptr<float>(), istd.data_ptr<float>(),
-                         has_affine ? gamma.data_ptr<float>() : nullptr,
-                         has_affine ? beta.data_ptr<float>() : nullptr,
+                         has_affine ? gamma.data_ptr<scalar_t>() : nullptr,
+                         has_affine ? beta.data_ptr<scalar_t>() : nullptr,
                          out.data_ptr<scalar_t>(), C, HW, relu ? 1 : 0,
                          has_affine ? 1 : 0);
     };
@@ -1309,13 +1360,22 @@ void bn_bwd_apply(Tensor x, Tensor dout, Tensor out, Tensor mean, Tensor istd,
                          dim3(threads), 0, cur_stream(), x.data_ptr<scalar_t>(),
                          dout.data_ptr<scalar_t>(), out.data_ptr<scalar_t>(),
                          mean.data_ptr<float>(), istd.data_ptr<float>(),
-                         has_affine ? gamma.data_ptr<float>() : nullptr,
+                         has_affine ? gamma.data_ptr<scalar_t>() : nullptr,
                          sums.data_ptr<float>(), dx.data_ptr<scalar_t>(), C, HW,
                          1.0f / (float)M, relu ? 1 : 0, has_affine ? 1 : 0,
                          use_batch ? 1 : 0);
     };
     if (vec) lp(std::true_type{}); else lp(std::false_type{});
   });
+}
+
+void ema_update(Tensor rm, Tensor a, Tensor rv, Tensor b, double momentum) {
+  const int64_t n1 = rm.numel(), n2 = rv.numel();
+  hipLaunchKernelGGL(dwt::ema_update_kernel,
+                     dim3(elementwise_blocks(n1 + n2, 256)), dim3(256), 0,
+                     cur_stream(), rm.data_ptr<float>(), a.data_ptr<float>(),
+                     n1, rv.data_ptr<float>(), b.data_ptr<float>(), n2,
+                     (float)momentum);
 }
 
 // ---------------------------- losses -------------------------------------
@@ -1375,6 +1435,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("bn_apply", &bn_apply);
   m.def("bn_bwd_reduce", &bn_bwd_reduce);
   m.def("bn_bwd_apply", &bn_bwd_apply);
+  m.def("ema_update", &ema_update);
   m.def("mec_fwd", &mec_fwd);
   m.def("mec_bwd", &mec_bwd);
   m.def("entropy_fwd", &entropy_fwd);
