@@ -73,7 +73,8 @@ def main():
     for name, param in model.named_parameters():
         (final_layer if name.startswith("fc_out") else rest).append(param)
     lr = 1e-2
-    optimizer = torch.optim.SGD(
+    from dwt_amd.ops.optim import FusedSGD
+    optimizer = FusedSGD(
         [{"params": rest}, {"params": final_layer, "lr": lr}],
         lr=lr * 0.1, momentum=0.9, weight_decay=5e-4)
 

@@ -910,6 +910,71 @@ __global__ void ema_update_kernel(float* __restrict__ rm,
 }
 
 // ===========================================================================
+// Fused multi-tensor optimizers.
+// One launch per param-group: grid.x indexes chunk descriptors
+// [p_ptr, g_ptr, state1_ptr, state2_ptr(or master), n].  bf16 params keep an
+// fp32 master copy (read-modify-write in fp32, bf16 shadow written back);
+// the kernel optionally zeroes the grad in the same pass so grad storage
+// stays stable across steps (no per-step re-zeroing launches).
+// Semantics match torch.optim.SGD (momentum, dampening=0, nesterov=False,
+// wd adds to grad) and torch.optim.Adam (classic, wd adds to grad).
+// ===========================================================================
+
+template <typename T, bool MASTER>
+__global__ void sgd_kernel(const int64_t* __restrict__ desc, float lr,
+                           float momentum, float wd, int zero_grad,
+                           int first_step) {
+  const int64_t* d = desc + (int64_t)blockIdx.x * 5;
+  T* p = reinterpret_cast<T*>(d[0]);
+  T* g = reinterpret_cast<T*>(d[1]);
+  float* mbuf = reinterpret_cast<float*>(d[2]);
+  float* master = reinterpret_cast<float*>(d[3]);
+  const int64_t n = d[4];
+  for (int64_t i = threadIdx.x; i < n; i += blockDim.x) {
+    float gv = ldf(g + i);
+    float pv = MASTER ? master[i] : ldf(p + i);
+    gv += wd * pv;
+    float m = gv;
+    if (momentum != 0.f) {
+      m = first_step ? gv : mbuf[i] * momentum + gv;
+      mbuf[i] = m;
+    }
+    pv -= lr * m;
+    if (MASTER) master[i] = pv;
+    stf(p + i, pv);
+    if (zero_grad) stf(g + i, 0.f);
+  }
+}
+
+template <typename T, bool MASTER>
+__global__ void adam_kernel(const int64_t* __restrict__ desc, float lr,
+                            float beta1, float beta2, float eps, float wd,
+                            float bc1, float bc2, int zero_grad) {
+  const int64_t* d = desc + (int64_t)blockIdx.x * 6;
+  T* p = reinterpret_cast<T*>(d[0]);
+  T* g = reinterpret_cast<T*>(d[1]);
+  float* mbuf = reinterpret_cast<float*>(d[2]);
+  float* vbuf = reinterpret_cast<float*>(d[3]);
+  float* master = reinterpret_cast<float*>(d[4]);
+  const int64_t n = d[5];
+  for (int64_t i = threadIdx.x; i < n; i += blockDim.x) {
+    float gv = ldf(g + i);
+    float pv = MASTER ? master[i] : ldf(p + i);
+    gv += wd * pv;
+    const float m = beta1 * mbuf[i] + (1.f - beta1) * gv;
+    const float v = beta2 * vbuf[i] + (1.f - beta2) * gv * gv;
+    mbuf[i] = m;
+    vbuf[i] = v;
+    const float mhat = m / bc1;
+    const float vhat = v / bc2;
+    pv -= lr * mhat / (sqrtf(vhat) + eps);
+    if (MASTER) master[i] = pv;
+    stf(p + i, pv);
+    if (zero_grad) stf(g + i, 0.f);
+  }
+}
+
+// ===========================================================================
 // Losses (fp32 logits, one wavefront per row)
 // ===========================================================================
 
@@ -1369,6 +1434,46 @@ void bn_bwd_apply(Tensor x, Tensor dout, Tensor out, Tensor mean, Tensor istd,
   });
 }
 
+void fused_sgd(Tensor desc, int64_t n_chunks, double lr, double momentum,
+               double wd, bool bf16_params, bool master, bool zero_grad,
+               bool first_step) {
+  auto lp = [&](auto tconst, auto mconst) {
+    using T = typename decltype(tconst)::type;
+    constexpr bool M = decltype(mconst)::value;
+    hipLaunchKernelGGL((dwt::sgd_kernel<T, M>), dim3(n_chunks), dim3(256), 0,
+                       cur_stream(), desc.data_ptr<int64_t>(), (float)lr,
+                       (float)momentum, (float)wd, zero_grad ? 1 : 0,
+                       first_step ? 1 : 0);
+  };
+  struct FT { using type = float; };
+  struct BT { using type = c10::BFloat16; };
+  if (bf16_params) {
+    if (master) lp(BT{}, std::true_type{}); else lp(BT{}, std::false_type{});
+  } else {
+    if (master) lp(FT{}, std::true_type{}); else lp(FT{}, std::false_type{});
+  }
+}
+
+void fused_adam(Tensor desc, int64_t n_chunks, double lr, double beta1,
+                double beta2, double eps, double wd, double bc1, double bc2,
+                bool bf16_params, bool master, bool zero_grad) {
+  auto lp = [&](auto tconst, auto mconst) {
+    using T = typename decltype(tconst)::type;
+    constexpr bool M = decltype(mconst)::value;
+    hipLaunchKernelGGL((dwt::adam_kernel<T, M>), dim3(n_chunks), dim3(256), 0,
+                       cur_stream(), desc.data_ptr<int64_t>(), (float)lr,
+                       (float)beta1, (float)beta2, (float)eps, (float)wd,
+                       (float)bc1, (float)bc2, zero_grad ? 1 : 0);
+  };
+  struct FT { using type = float; };
+  struct BT { using type = c10::BFloat16; };
+  if (bf16_params) {
+    if (master) lp(BT{}, std::true_type{}); else lp(BT{}, std::false_type{});
+  } else {
+    if (master) lp(FT{}, std::true_type{}); else lp(FT{}, std::false_type{});
+  }
+}
+
 void ema_update(Tensor rm, Tensor a, Tensor rv, Tensor b, double momentum) {
   const int64_t n1 = rm.numel(), n2 = rv.numel();
   hipLaunchKernelGGL(dwt::ema_update_kernel,
@@ -1436,6 +1541,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("bn_bwd_reduce", &bn_bwd_reduce);
   m.def("bn_bwd_apply", &bn_bwd_apply);
   m.def("ema_update", &ema_update);
+  m.def("fused_sgd", &fused_sgd);
+  m.def("fused_adam", &fused_adam);
   m.def("mec_fwd", &mec_fwd);
   m.def("mec_bwd", &mec_bwd);
   m.def("entropy_fwd", &entropy_fwd);

@@ -131,7 +131,8 @@ def main(argv=None):
     final_layer_params, rest = [], []
     for name, param in model.named_parameters():
         (final_layer_params if name.startswith('fc_out') else rest).append(param)
-    optimizer = optim.SGD([
+    from dwt_amd.ops.optim import FusedSGD
+    optimizer = FusedSGD([
         {'params': rest},
         {'params': final_layer_params, 'lr': args.lr},
     ], lr=args.lr * 0.1, momentum=0.9, weight_decay=5e-4)

@@ -123,7 +123,8 @@ def main(argv=None):
     model = LeNet(group_size=args.group_size,
                   streams=3 if args.loss == 'mec' else 2,
                   whiten_mode=args.whiten_mode).to(device).to(dtype)
-    optimizer = optim.Adam(model.parameters(), lr=args.lr, weight_decay=5e-4)
+    from dwt_amd.ops.optim import FusedAdam
+    optimizer = FusedAdam(model.parameters(), lr=args.lr, weight_decay=5e-4)
     sched = lr_scheduler.MultiStepLR(optimizer, milestones=[50, 80], gamma=0.1)
 
     start_epoch = 0
