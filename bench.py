@@ -37,6 +37,8 @@ def get_args():
     p.add_argument("--whiten_mode", choices=["chol", "zca"], default="chol")
     p.add_argument("--device", default=None, help="cpu fallback for tests")
     p.add_argument("--layers", default="3,4,6,3")
+    p.add_argument("--channels_last", type=int, default=1,
+                   help="NHWC layout (MIOpen-native on gfx950; 0 = NCHW)")
     return p.parse_args()
 
 
@@ -68,6 +70,9 @@ def main():
     model = ResNetDWT(Bottleneck, layers, None, num_classes=args.num_classes,
                       group_size=args.group_size, whiten_mode=args.whiten_mode)
     model = model.to(device).to(dtype).train()
+    use_cl = bool(args.channels_last) and use_cuda
+    if use_cl:
+        model = model.to(memory_format=torch.channels_last)
 
     final_layer, rest = [], []
     for name, param in model.named_parameters():
@@ -83,6 +88,8 @@ def main():
     b = args.batch
     n_per_gpu = 3 * b
     data = torch.randn(n_per_gpu, 3, args.img, args.img, device=device, dtype=dtype)
+    if use_cl:
+        data = data.contiguous(memory_format=torch.channels_last)
     labels = torch.randint(0, args.num_classes, (b,), device=device)
 
     def step():

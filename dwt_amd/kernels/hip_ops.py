@@ -19,6 +19,25 @@ _FAST_G = (2, 4, 8)
 _warned = set()
 
 
+def _whiten_layout(x, c):
+    """'cl' when x is channels_last and the NHWC kernels support C."""
+    if x.dim() == 4 and x.is_contiguous(memory_format=torch.channels_last) \
+            and not x.is_contiguous():
+        if c % 64 == 0 and (c <= 256 or c % 256 == 0):
+            return "cl"
+    return "nchw"
+
+
+def _bn_layout(x, c):
+    if x.dim() == 2 and c % 4 == 0 and x.is_contiguous():
+        return "cl"  # (N, C) IS channels-last: C contiguous per position
+    if x.dim() == 4 and x.is_contiguous(memory_format=torch.channels_last) \
+            and not x.is_contiguous():
+        if c % 4 == 0 and (c <= 1024 or c % 1024 == 0):
+            return "cl"
+    return "nchw"
+
+
 def _warn_once(key, msg):
     if key not in _warned:
         _warned.add(key)
@@ -45,7 +64,10 @@ class _HipWhitenMulti(torch.autograd.Function):
         ns_iters = cfg.get("ns_iters", 7)
         use_batch = training or not track
 
-        x = x.contiguous()
+        c = x.shape[1]
+        layout = _whiten_layout(x, c)
+        if layout == "nchw":
+            x = x.contiguous()
         n, c, h, w = x.shape
         b = n // parts
         n_groups = c // g
@@ -64,7 +86,10 @@ class _HipWhitenMulti(torch.autograd.Function):
                 acc = torch.zeros(n_groups * (g + g * g), device=dev, dtype=torch.float32)
                 mean = torch.empty(c, device=dev, dtype=torch.float32)
                 cov = torch.empty(n_groups, g, g, device=dev, dtype=torch.float32)
-                ext.whiten_stats(xp, acc, mean, cov, g)
+                if layout == "cl":
+                    ext.whiten_stats_cl(xp, acc, mean, cov, g, c, b * h * w)
+                else:
+                    ext.whiten_stats(xp, acc, mean, cov, g)
             else:
                 mean = _f32c(running_means[p]).reshape(c)
                 cov = _f32c(running_vars[p]).reshape(n_groups, g, g)
@@ -81,8 +106,13 @@ class _HipWhitenMulti(torch.autograd.Function):
                 ext.matfn_ns_fwd(cov, wmat, ys, zs, svals, eps, ns_iters)
                 saved_mat.append((ys, zs, svals))
 
-            ext.whiten_apply(xp, mean, wmat, gflat, bflat,
-                             out[p * b:(p + 1) * b], g, relu, has_affine)
+            if layout == "cl":
+                ext.whiten_apply_cl(xp, mean, wmat, gflat, bflat,
+                                    out[p * b:(p + 1) * b], g, c, b * h * w,
+                                    relu, has_affine)
+            else:
+                ext.whiten_apply(xp, mean, wmat, gflat, bflat,
+                                 out[p * b:(p + 1) * b], g, relu, has_affine)
             means.append(mean)
             wmats.append(wmat)
 
@@ -100,6 +130,7 @@ class _HipWhitenMulti(torch.autograd.Function):
 
         ctx.cfg = cfg
         ctx.g = g
+        ctx.layout = layout
         ctx.m_count = m_count
         ctx.means = means
         ctx.wmats = wmats
@@ -124,7 +155,11 @@ class _HipWhitenMulti(torch.autograd.Function):
         b = n // parts
         n_groups = c // g
         dev = x.device
-        dout = dout.contiguous()
+        layout = ctx.layout
+        if layout == "cl":
+            dout = dout.contiguous(memory_format=torch.channels_last)
+        else:
+            dout = dout.contiguous()
 
         dgb = torch.zeros(parts, 2, c, device=dev, dtype=torch.float32)
         dx = torch.empty_like(x)
@@ -133,8 +168,13 @@ class _HipWhitenMulti(torch.autograd.Function):
             xp, doutp, outp = x[sl], dout[sl], out[sl]
             mean, wmat = ctx.means[p], ctx.wmats[p]
             dW = torch.zeros(n_groups, g, g, device=dev, dtype=torch.float32)
-            ext.whiten_bwd_reduce(xp, doutp, outp, mean, wmat, ctx.gflat,
-                                  dW, dgb[p].reshape(-1), g, relu, ctx.has_affine)
+            if layout == "cl":
+                ext.whiten_bwd_reduce_cl(xp, doutp, outp, mean, wmat, ctx.gflat,
+                                         dW, dgb[p].reshape(-1), g, c, b * h * w,
+                                         relu, ctx.has_affine)
+            else:
+                ext.whiten_bwd_reduce(xp, doutp, outp, mean, wmat, ctx.gflat,
+                                      dW, dgb[p].reshape(-1), g, relu, ctx.has_affine)
             if use_batch:
                 gdb = (ctx.gflat.float() * dgb[p, 1]) if ctx.has_affine else dgb[p, 1]
                 gdb = gdb.contiguous()
@@ -151,9 +191,14 @@ class _HipWhitenMulti(torch.autograd.Function):
             else:
                 S = torch.empty(0, device=dev)
                 corr = torch.empty(0, device=dev)
-            ext.whiten_bwd_apply(xp, doutp, outp, mean, wmat, ctx.gflat,
-                                 S, corr, dx[sl], g, relu, ctx.has_affine,
-                                 use_batch)
+            if layout == "cl":
+                ext.whiten_bwd_apply_cl(xp, doutp, outp, mean, wmat, ctx.gflat,
+                                        S, corr, dx[sl], g, c, b * h * w, relu,
+                                        ctx.has_affine, use_batch)
+            else:
+                ext.whiten_bwd_apply(xp, doutp, outp, mean, wmat, ctx.gflat,
+                                     S, corr, dx[sl], g, relu, ctx.has_affine,
+                                     use_batch)
 
         if ctx.has_affine:
             dgamma = dgb[:, 0].sum(0).reshape(gamma.shape).to(gamma.dtype)
@@ -185,8 +230,11 @@ class _HipBatchNormMulti(torch.autograd.Function):
         use_batch = training or not track
 
         spatial = x.dim() == 4
-        x = x.contiguous()
-        n, c = x.shape[0], x.shape[1]
+        c = x.shape[1]
+        layout = _bn_layout(x, c)
+        if layout == "nchw" or x.dim() == 2:
+            x = x.contiguous()
+        n = x.shape[0]
         b = n // parts
         dev = x.device
         cnt = (x.numel() // parts) // c
@@ -204,7 +252,10 @@ class _HipBatchNormMulti(torch.autograd.Function):
                 mean = torch.empty(c, device=dev, dtype=torch.float32)
                 istd = torch.empty(c, device=dev, dtype=torch.float32)
                 var_unb = torch.empty(c, device=dev, dtype=torch.float32)
-                ext.bn_stats(xp, acc, mean, istd, var_unb, eps)
+                if layout == "cl":
+                    ext.bn_stats_cl(xp, acc, mean, istd, var_unb, c, cnt, eps)
+                else:
+                    ext.bn_stats(xp, acc, mean, istd, var_unb, eps)
                 if training and track and running_means is not None:
                     with torch.no_grad():
                         rm, rv = running_means[p], running_vars[p]
@@ -218,12 +269,18 @@ class _HipBatchNormMulti(torch.autograd.Function):
                 mean = _f32c(running_means[p]).reshape(c)
                 var = _f32c(running_vars[p]).reshape(c)
                 istd = torch.rsqrt(var + eps)
-            ext.bn_apply(xp, mean, istd, gflat, bflat,
-                         out[p * b:(p + 1) * b], relu, has_affine)
+            if layout == "cl":
+                ext.bn_apply_cl(xp, mean, istd, gflat, bflat,
+                                out[p * b:(p + 1) * b], c, cnt, relu, has_affine)
+            else:
+                ext.bn_apply(xp, mean, istd, gflat, bflat,
+                             out[p * b:(p + 1) * b], relu, has_affine)
             means.append(mean)
             istds.append(istd)
 
         ctx.cfg = cfg
+        ctx.layout = layout
+        ctx.cnt = cnt
         ctx.spatial = spatial
         ctx.means = means
         ctx.istds = istds
@@ -242,18 +299,29 @@ class _HipBatchNormMulti(torch.autograd.Function):
         n, c = x.shape[0], x.shape[1]
         b = n // parts
         dev = x.device
-        dout = dout.contiguous()
+        layout = ctx.layout
+        if layout == "cl" and x.dim() == 4:
+            dout = dout.contiguous(memory_format=torch.channels_last)
+        else:
+            dout = dout.contiguous()
 
         sums = torch.zeros(parts, 2, c, device=dev, dtype=torch.float32)
         dx = torch.empty_like(x)
         for p in range(parts):
             sl = slice(p * b, (p + 1) * b)
             xp, doutp, outp = x[sl], dout[sl], out[sl]
-            ext.bn_bwd_reduce(xp, doutp, outp, ctx.means[p], ctx.istds[p],
-                              sums[p].reshape(-1), relu)
-            ext.bn_bwd_apply(xp, doutp, outp, ctx.means[p], ctx.istds[p],
-                             ctx.gflat, sums[p].reshape(-1), dx[sl], relu,
-                             ctx.has_affine, ctx.use_batch)
+            if layout == "cl":
+                ext.bn_bwd_reduce_cl(xp, doutp, outp, ctx.means[p], ctx.istds[p],
+                                     sums[p].reshape(-1), c, ctx.cnt, relu)
+                ext.bn_bwd_apply_cl(xp, doutp, outp, ctx.means[p], ctx.istds[p],
+                                    ctx.gflat, sums[p].reshape(-1), dx[sl], c,
+                                    ctx.cnt, relu, ctx.has_affine, ctx.use_batch)
+            else:
+                ext.bn_bwd_reduce(xp, doutp, outp, ctx.means[p], ctx.istds[p],
+                                  sums[p].reshape(-1), relu)
+                ext.bn_bwd_apply(xp, doutp, outp, ctx.means[p], ctx.istds[p],
+                                 ctx.gflat, sums[p].reshape(-1), dx[sl], relu,
+                                 ctx.has_affine, ctx.use_batch)
 
         if ctx.has_affine:
             # dgamma = sum dy * xhat ; dbeta = sum dy  (sums[:,1] is dy*xhat)

@@ -725,6 +725,442 @@ __global__ void whiten_bwd_apply_kernel(
 }
 
 // ===========================================================================
+// NHWC (channels_last) variants.
+//
+// Layout x[n, h, w, c] with c fastest: the group's g channels are CONTIGUOUS
+// bytes, so one lane loads its whole group as one 8/16-B access and lanes
+// tile (position, group) space fully coalesced — the layout MIOpen's bf16
+// igemm kernels want anyway (running channels_last removes every
+// batched_transpose around the convs, measured ~13% of step time).
+//
+// Thread mapping for the reduce kernels (fixed group per thread so the
+// g x g accumulators live in registers):
+//   CW = min(C, 256) channels per z-slice, GW = CW/g groups per slice,
+//   thread t handles group (t % GW), position rows advance by 256/GW.
+//   Requires 256 % GW == 0 (C in {64,128,256,512,1024,2048} with g 2/4/8).
+// ===========================================================================
+
+template <typename T, int G>
+DEV_INLINE void load_group(const T* p, float* out);
+
+template <> DEV_INLINE void load_group<float, 2>(const float* p, float* o) {
+  const float2 v = *reinterpret_cast<const float2*>(p);
+  o[0] = v.x; o[1] = v.y;
+}
+template <> DEV_INLINE void load_group<float, 4>(const float* p, float* o) {
+  const float4 v = *reinterpret_cast<const float4*>(p);
+  o[0] = v.x; o[1] = v.y; o[2] = v.z; o[3] = v.w;
+}
+template <> DEV_INLINE void load_group<float, 8>(const float* p, float* o) {
+  load_group<float, 4>(p, o);
+  load_group<float, 4>(p + 4, o + 4);
+}
+DEV_INLINE void bf16_to_f32x2(unsigned int w, float* o) {
+  union { unsigned int u; float f; } lo, hi;
+  lo.u = (w & 0xffffu) << 16; hi.u = w & 0xffff0000u;
+  o[0] = lo.f; o[1] = hi.f;
+}
+template <> DEV_INLINE void load_group<c10::BFloat16, 2>(const c10::BFloat16* p, float* o) {
+  bf16_to_f32x2(*reinterpret_cast<const unsigned int*>(p), o);
+}
+template <> DEV_INLINE void load_group<c10::BFloat16, 4>(const c10::BFloat16* p, float* o) {
+  const uint2 v = *reinterpret_cast<const uint2*>(p);
+  bf16_to_f32x2(v.x, o); bf16_to_f32x2(v.y, o + 2);
+}
+template <> DEV_INLINE void load_group<c10::BFloat16, 8>(const c10::BFloat16* p, float* o) {
+  const uint4 v = *reinterpret_cast<const uint4*>(p);
+  bf16_to_f32x2(v.x, o); bf16_to_f32x2(v.y, o + 2);
+  bf16_to_f32x2(v.z, o + 4); bf16_to_f32x2(v.w, o + 6);
+}
+
+DEV_INLINE unsigned int f32x2_to_bf16(float a, float b) {
+  union { unsigned int u; float f; } x, y;
+  x.f = a; y.f = b;
+  unsigned int la = (x.u >> 16) & 1u, lb = (y.u >> 16) & 1u;
+  unsigned int ra = (x.u + 0x7fffu + la) >> 16;
+  unsigned int rb = (y.u + 0x7fffu + lb) >> 16;
+  return (ra & 0xffffu) | (rb << 16);
+}
+template <typename T, int G>
+DEV_INLINE void store_group(T* p, const float* in);
+template <> DEV_INLINE void store_group<float, 2>(float* p, const float* i) {
+  *reinterpret_cast<float2*>(p) = make_float2(i[0], i[1]);
+}
+template <> DEV_INLINE void store_group<float, 4>(float* p, const float* i) {
+  *reinterpret_cast<float4*>(p) = make_float4(i[0], i[1], i[2], i[3]);
+}
+template <> DEV_INLINE void store_group<float, 8>(float* p, const float* i) {
+  store_group<float, 4>(p, i); store_group<float, 4>(p + 4, i + 4);
+}
+template <> DEV_INLINE void store_group<c10::BFloat16, 2>(c10::BFloat16* p, const float* i) {
+  *reinterpret_cast<unsigned int*>(p) = f32x2_to_bf16(i[0], i[1]);
+}
+template <> DEV_INLINE void store_group<c10::BFloat16, 4>(c10::BFloat16* p, const float* i) {
+  *reinterpret_cast<uint2*>(p) = make_uint2(f32x2_to_bf16(i[0], i[1]),
+                                            f32x2_to_bf16(i[2], i[3]));
+}
+template <> DEV_INLINE void store_group<c10::BFloat16, 8>(c10::BFloat16* p, const float* i) {
+  *reinterpret_cast<uint4*>(p) =
+      make_uint4(f32x2_to_bf16(i[0], i[1]), f32x2_to_bf16(i[2], i[3]),
+                 f32x2_to_bf16(i[4], i[5]), f32x2_to_bf16(i[6], i[7]));
+}
+
+// fused mean+cov, NHWC
+template <typename T, int G>
+__global__ void whiten_stats_nhwc_kernel(
+    const T* __restrict__ x, float* __restrict__ acc,
+    int C, int64_t M /* = N*H*W positions */) {
+  const int CW = C < 256 ? C : 256;
+  const int GW = CW / G;
+  const int c0 = blockIdx.z * 256 + (threadIdx.x % GW) * G;
+  const int grp = c0 / G;
+  const int rows_per_iter = blockDim.x / GW;
+  const int row_in_block = threadIdx.x / GW;
+
+  float s[G], p[G][G];
+#pragma unroll
+  for (int i = 0; i < G; ++i) {
+    s[i] = 0.f;
+#pragma unroll
+    for (int j = 0; j < G; ++j) p[i][j] = 0.f;
+  }
+  const int64_t stride = (int64_t)gridDim.x * rows_per_iter;
+  for (int64_t m = (int64_t)blockIdx.x * rows_per_iter + row_in_block; m < M;
+       m += stride) {
+    float v[G];
+    load_group<T, G>(x + m * C + c0, v);
+#pragma unroll
+    for (int i = 0; i < G; ++i) {
+      s[i] += v[i];
+#pragma unroll
+      for (int j = 0; j <= i; ++j) p[i][j] += v[i] * v[j];
+    }
+  }
+  // reduce within the wave over lanes sharing the same group (stride GW in
+  // lane space is irregular) — go through LDS per-block instead: each value
+  // atomically added into a per-block LDS accumulator, then one global
+  // atomic per value.
+  constexpr int NTRI = G * (G + 1) / 2;
+  __shared__ float lacc[(256 / G) * (G + NTRI)];  // [GW][G+NTRI], GW <= 256/G
+  float* mine = lacc + (threadIdx.x % GW) * (G + NTRI);
+  // zero once
+  for (int k = threadIdx.x; k < GW * (G + NTRI); k += blockDim.x) {
+    lacc[k] = 0.f;
+  }
+  __syncthreads();
+#pragma unroll
+  for (int i = 0; i < G; ++i) atomicAdd(&mine[i], s[i]);
+#pragma unroll
+  for (int i = 0; i < G; ++i)
+#pragma unroll
+    for (int j = 0; j <= i; ++j)
+      atomicAdd(&mine[G + i * (i + 1) / 2 + j], p[i][j]);
+  __syncthreads();
+  // first GW*(G+NTRI) threads flush to global
+  for (int k = threadIdx.x; k < GW * (G + NTRI); k += blockDim.x) {
+    const int lg = k / (G + NTRI);
+    const int kk = k % (G + NTRI);
+    const int gidx = blockIdx.z * (256 / G) + lg;
+    float* gacc = acc + (int64_t)gidx * (G + G * G);
+    if (kk < G) {
+      atomicAdd(&gacc[kk], lacc[k]);
+    } else {
+      int rem = kk - G, i = 0;
+      while (rem > i) { rem -= (i + 1); ++i; }
+      atomicAdd(&gacc[G + i * G + rem], lacc[k]);
+    }
+  }
+}
+
+// fused apply, NHWC: lanes tile (position, group) flat
+template <typename T, int G>
+__global__ void whiten_apply_nhwc_kernel(
+    const T* __restrict__ x, const float* __restrict__ mean,
+    const float* __restrict__ W, const T* __restrict__ gamma,
+    const T* __restrict__ beta, T* __restrict__ out,
+    int C, int64_t total /* = M * n_groups */, int relu, int has_affine) {
+  const int n_groups = C / G;
+  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t f = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; f < total;
+       f += stride) {
+    const int grp = (int)(f % n_groups);
+    const int64_t m = f / n_groups;
+    const int c0 = grp * G;
+    float v[G], y[G];
+    load_group<T, G>(x + m * C + c0, v);
+    const float* Wg = W + (int64_t)grp * G * G;
+#pragma unroll
+    for (int j = 0; j < G; ++j) v[j] -= mean[c0 + j];
+#pragma unroll
+    for (int i = 0; i < G; ++i) {
+      float a = 0.f;
+#pragma unroll
+      for (int j = 0; j < G; ++j) a += Wg[i * G + j] * v[j];
+      if (has_affine) a = a * ldf(gamma + c0 + i) + ldf(beta + c0 + i);
+      y[i] = relu ? fmaxf(a, 0.f) : a;
+    }
+    store_group<T, G>(out + m * C + c0, y);
+  }
+}
+
+// backward reduce, NHWC (same mapping as stats)
+template <typename T, int G>
+__global__ void whiten_bwd_reduce_nhwc_kernel(
+    const T* __restrict__ x, const T* __restrict__ dout,
+    const T* __restrict__ out, const float* __restrict__ mean,
+    const float* __restrict__ W, const T* __restrict__ gamma,
+    float* __restrict__ dWacc, float* __restrict__ dgb,
+    int C, int64_t M, int relu, int has_affine) {
+  const int CW = C < 256 ? C : 256;
+  const int GW = CW / G;
+  const int c0 = blockIdx.z * 256 + (threadIdx.x % GW) * G;
+  const int grp = c0 / G;
+  const int rows_per_iter = blockDim.x / GW;
+  const int row_in_block = threadIdx.x / GW;
+
+  float m_[G], Wr[G][G], gm[G];
+#pragma unroll
+  for (int i = 0; i < G; ++i) {
+    m_[i] = mean[c0 + i];
+    gm[i] = has_affine ? ldf(gamma + c0 + i) : 1.f;
+#pragma unroll
+    for (int j = 0; j < G; ++j) Wr[i][j] = W[((int64_t)grp * G + i) * G + j];
+  }
+  float dWl[G][G], dg[G], db[G];
+#pragma unroll
+  for (int i = 0; i < G; ++i) {
+    dg[i] = 0.f; db[i] = 0.f;
+#pragma unroll
+    for (int j = 0; j < G; ++j) dWl[i][j] = 0.f;
+  }
+  const int64_t stride = (int64_t)gridDim.x * rows_per_iter;
+  for (int64_t m = (int64_t)blockIdx.x * rows_per_iter + row_in_block; m < M;
+       m += stride) {
+    float xv[G], dyv[G], ov[G];
+    load_group<T, G>(x + m * C + c0, xv);
+    load_group<T, G>(dout + m * C + c0, dyv);
+    if (relu) {
+      load_group<T, G>(out + m * C + c0, ov);
+#pragma unroll
+      for (int j = 0; j < G; ++j) dyv[j] = ov[j] > 0.f ? dyv[j] : 0.f;
+    }
+#pragma unroll
+    for (int j = 0; j < G; ++j) xv[j] -= m_[j];
+#pragma unroll
+    for (int i = 0; i < G; ++i) {
+      const float dy = dyv[i];
+      db[i] += dy;
+      float y0 = 0.f;
+#pragma unroll
+      for (int j = 0; j < G; ++j) y0 += Wr[i][j] * xv[j];
+      dg[i] += dy * y0;
+      const float dy0 = dy * gm[i];
+#pragma unroll
+      for (int j = 0; j < G; ++j) dWl[i][j] += dy0 * xv[j];
+    }
+  }
+  // per-block LDS accumulate by group slot, then one global atomic per value
+  constexpr int NV = 2 * G + G * G;
+  __shared__ float lacc[(256 / G) * NV];  // [GW][NV], GW <= 256/G
+  float* mine = lacc + (threadIdx.x % GW) * NV;
+  for (int k = threadIdx.x; k < GW * NV; k += blockDim.x) lacc[k] = 0.f;
+  __syncthreads();
+#pragma unroll
+  for (int i = 0; i < G; ++i) {
+    atomicAdd(&mine[i], dg[i]);
+    atomicAdd(&mine[G + i], db[i]);
+#pragma unroll
+    for (int j = 0; j < G; ++j) atomicAdd(&mine[2 * G + i * G + j], dWl[i][j]);
+  }
+  __syncthreads();
+  for (int k = threadIdx.x; k < GW * NV; k += blockDim.x) {
+    const int lg = k / NV;
+    const int kk = k % NV;
+    const int gidx = blockIdx.z * (256 / G) + lg;
+    const int cg0 = gidx * G;
+    if (kk < G) atomicAdd(&dgb[cg0 + kk], lacc[k]);
+    else if (kk < 2 * G) atomicAdd(&dgb[C + cg0 + (kk - G)], lacc[k]);
+    else atomicAdd(&dWacc[(int64_t)gidx * G * G + (kk - 2 * G)], lacc[k]);
+  }
+}
+
+// backward apply, NHWC
+template <typename T, int G>
+__global__ void whiten_bwd_apply_nhwc_kernel(
+    const T* __restrict__ x, const T* __restrict__ dout,
+    const T* __restrict__ out, const float* __restrict__ mean,
+    const float* __restrict__ W, const T* __restrict__ gamma,
+    const float* __restrict__ S, const float* __restrict__ corr,
+    T* __restrict__ dx, int C, int64_t total, int relu, int has_affine,
+    int train_stats) {
+  const int n_groups = C / G;
+  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t f = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; f < total;
+       f += stride) {
+    const int grp = (int)(f % n_groups);
+    const int64_t m = f / n_groups;
+    const int c0 = grp * G;
+    float xv[G], dyv[G], ov[G], r[G];
+    load_group<T, G>(x + m * C + c0, xv);
+    load_group<T, G>(dout + m * C + c0, dyv);
+    if (relu) {
+      load_group<T, G>(out + m * C + c0, ov);
+#pragma unroll
+      for (int j = 0; j < G; ++j) dyv[j] = ov[j] > 0.f ? dyv[j] : 0.f;
+    }
+    const float* Wg = W + (int64_t)grp * G * G;
+    const float* Sg = S + (int64_t)grp * G * G;
+#pragma unroll
+    for (int j = 0; j < G; ++j) {
+      xv[j] -= mean[c0 + j];
+      if (has_affine) dyv[j] *= ldf(gamma + c0 + j);
+    }
+#pragma unroll
+    for (int i = 0; i < G; ++i) {
+      float a = train_stats ? -corr[c0 + i] : 0.f;
+#pragma unroll
+      for (int j = 0; j < G; ++j) {
+        a += Wg[j * G + i] * dyv[j];
+        if (train_stats) a += Sg[i * G + j] * xv[j];
+      }
+      r[i] = a;
+    }
+    store_group<T, G>(dx + m * C + c0, r);
+  }
+}
+
+// BN NHWC: per-lane 4-channel chunks
+template <typename T>
+__global__ void bn_stats_nhwc_kernel(
+    const T* __restrict__ x, float* __restrict__ acc, int C, int64_t M) {
+  constexpr int VC = 4;
+  const int CW = C < 1024 ? C : 1024;
+  const int NCH = CW / VC;
+  const int c0 = blockIdx.z * 1024 + (threadIdx.x % NCH) * VC;
+  const int rows_per_iter = blockDim.x / NCH;
+  const int row_in_block = threadIdx.x / NCH;
+  float s[VC], ss[VC];
+#pragma unroll
+  for (int k = 0; k < VC; ++k) { s[k] = 0.f; ss[k] = 0.f; }
+  const int64_t stride = (int64_t)gridDim.x * rows_per_iter;
+  for (int64_t m = (int64_t)blockIdx.x * rows_per_iter + row_in_block; m < M;
+       m += stride) {
+    float v[VC];
+    load_group<T, VC>(x + m * C + c0, v);
+#pragma unroll
+    for (int k = 0; k < VC; ++k) { s[k] += v[k]; ss[k] += v[k] * v[k]; }
+  }
+#pragma unroll
+  for (int k = 0; k < VC; ++k) {
+    atomicAdd(&acc[c0 + k], s[k]);
+    atomicAdd(&acc[C + c0 + k], ss[k]);
+  }
+}
+
+template <typename T>
+__global__ void bn_apply_nhwc_kernel(
+    const T* __restrict__ x, const float* __restrict__ mean,
+    const float* __restrict__ istd, const T* __restrict__ gamma,
+    const T* __restrict__ beta, T* __restrict__ out, int C, int64_t total4,
+    int relu, int has_affine) {
+  constexpr int VC = 4;
+  const int nch = C / VC;
+  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t f = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; f < total4;
+       f += stride) {
+    const int c0 = (int)(f % nch) * VC;
+    const int64_t m = f / nch;
+    float v[VC];
+    load_group<T, VC>(x + m * C + c0, v);
+#pragma unroll
+    for (int k = 0; k < VC; ++k) {
+      float y = (v[k] - mean[c0 + k]) * istd[c0 + k];
+      if (has_affine) y = y * ldf(gamma + c0 + k) + ldf(beta + c0 + k);
+      v[k] = relu ? fmaxf(y, 0.f) : y;
+    }
+    store_group<T, VC>(out + m * C + c0, v);
+  }
+}
+
+template <typename T>
+__global__ void bn_bwd_reduce_nhwc_kernel(
+    const T* __restrict__ x, const T* __restrict__ dout,
+    const T* __restrict__ out, const float* __restrict__ mean,
+    const float* __restrict__ istd, float* __restrict__ sums, int C,
+    int64_t M, int relu) {
+  constexpr int VC = 4;
+  const int CW = C < 1024 ? C : 1024;
+  const int NCH = CW / VC;
+  const int c0 = blockIdx.z * 1024 + (threadIdx.x % NCH) * VC;
+  const int rows_per_iter = blockDim.x / NCH;
+  const int row_in_block = threadIdx.x / NCH;
+  float s_dy[VC], s_dyxh[VC];
+#pragma unroll
+  for (int k = 0; k < VC; ++k) { s_dy[k] = 0.f; s_dyxh[k] = 0.f; }
+  const int64_t stride = (int64_t)gridDim.x * rows_per_iter;
+  for (int64_t m = (int64_t)blockIdx.x * rows_per_iter + row_in_block; m < M;
+       m += stride) {
+    float xv[VC], dv[VC], ov[VC];
+    load_group<T, VC>(x + m * C + c0, xv);
+    load_group<T, VC>(dout + m * C + c0, dv);
+    if (relu) {
+      load_group<T, VC>(out + m * C + c0, ov);
+#pragma unroll
+      for (int k = 0; k < VC; ++k) dv[k] = ov[k] > 0.f ? dv[k] : 0.f;
+    }
+#pragma unroll
+    for (int k = 0; k < VC; ++k) {
+      s_dy[k] += dv[k];
+      s_dyxh[k] += dv[k] * (xv[k] - mean[c0 + k]) * istd[c0 + k];
+    }
+  }
+#pragma unroll
+  for (int k = 0; k < VC; ++k) {
+    atomicAdd(&sums[c0 + k], s_dy[k]);
+    atomicAdd(&sums[C + c0 + k], s_dyxh[k]);
+  }
+}
+
+template <typename T>
+__global__ void bn_bwd_apply_nhwc_kernel(
+    const T* __restrict__ x, const T* __restrict__ dout,
+    const T* __restrict__ out, const float* __restrict__ mean,
+    const float* __restrict__ istd, const T* __restrict__ gamma,
+    const float* __restrict__ sums, T* __restrict__ dx, int C, int64_t total4,
+    float inv_m, int relu, int has_affine, int use_batch) {
+  constexpr int VC = 4;
+  const int nch = C / VC;
+  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t f = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; f < total4;
+       f += stride) {
+    const int c0 = (int)(f % nch) * VC;
+    const int64_t m = f / nch;
+    float xv[VC], dv[VC], ov[VC];
+    load_group<T, VC>(x + m * C + c0, xv);
+    load_group<T, VC>(dout + m * C + c0, dv);
+    if (relu) {
+      load_group<T, VC>(out + m * C + c0, ov);
+#pragma unroll
+      for (int k = 0; k < VC; ++k) dv[k] = ov[k] > 0.f ? dv[k] : 0.f;
+    }
+#pragma unroll
+    for (int k = 0; k < VC; ++k) {
+      const int c = c0 + k;
+      const float gm = has_affine ? ldf(gamma + c) : 1.f;
+      const float is = istd[c];
+      const float dxh = dv[k] * gm;
+      if (use_batch) {
+        const float xh = (xv[k] - mean[c]) * is;
+        dv[k] = (dxh - gm * sums[c] * inv_m - xh * gm * sums[C + c] * inv_m) * is;
+      } else {
+        dv[k] = dxh * is;
+      }
+    }
+    store_group<T, VC>(dx + m * C + c0, dv);
+  }
+}
+
+// ===========================================================================
 // Domain BatchNorm
 // ===========================================================================
 
@@ -1173,6 +1609,170 @@ void whiten_stats(Tensor x, Tensor acc, Tensor mean, Tensor cov, int64_t g) {
                      (int)g, 1.0f / (float)M);
 }
 
+// ---------------------- NHWC (channels_last) launchers -------------------
+// x is the raw NHWC storage; M = N*H*W positions, C channels.
+
+#define DWT_SWITCH_G(g, body)                                                  \
+  switch (g) {                                                                 \
+    case 2: { constexpr int G = 2; body; break; }                              \
+    case 4: { constexpr int G = 4; body; break; }                              \
+    case 8: { constexpr int G = 8; body; break; }                              \
+    default: TORCH_CHECK(false, "unsupported group size ", g);                 \
+  }
+
+inline int64_t nhwc_reduce_blocks(int64_t M, int rows_per_iter, int zslices) {
+  const int64_t want = (M + (int64_t)rows_per_iter * 16 - 1) / ((int64_t)rows_per_iter * 16);
+  const int64_t cap = std::max<int64_t>(1, 4096 / std::max(zslices, 1));
+  return std::min(std::max<int64_t>(want, 1), cap);
+}
+
+void whiten_stats_cl(Tensor x, Tensor acc, Tensor mean, Tensor cov, int64_t g,
+                     int64_t C, int64_t M) {
+  const int zslices = (C + 255) / 256;
+  DISPATCH_FT(x, "whiten_stats_cl", [&] {
+    DWT_SWITCH_G(g, {
+      const int GW = (C < 256 ? C : 256) / G;
+      dim3 grid(nhwc_reduce_blocks(M, 256 / GW, zslices), 1, zslices);
+      hipLaunchKernelGGL((dwt::whiten_stats_nhwc_kernel<scalar_t, G>), grid,
+                         dim3(256), 0, cur_stream(), x.data_ptr<scalar_t>(),
+                         acc.data_ptr<float>(), (int)C, M);
+    });
+  });
+  const int n_groups = C / g;
+  const int fin_threads = std::min<int64_t>(g * g, 1024);
+  hipLaunchKernelGGL(dwt::whiten_stats_final_kernel, dim3(n_groups),
+                     dim3(fin_threads), 0, cur_stream(), acc.data_ptr<float>(),
+                     mean.data_ptr<float>(), cov.data_ptr<float>(), n_groups,
+                     (int)g, 1.0f / (float)M);
+}
+
+void whiten_apply_cl(Tensor x, Tensor mean, Tensor W, Tensor gamma, Tensor beta,
+                     Tensor out, int64_t g, int64_t C, int64_t M, bool relu,
+                     bool has_affine) {
+  const int64_t total = M * (C / g);
+  DISPATCH_FT(x, "whiten_apply_cl", [&] {
+    DWT_SWITCH_G(g, {
+      hipLaunchKernelGGL((dwt::whiten_apply_nhwc_kernel<scalar_t, G>),
+                         dim3(std::min<int64_t>(elementwise_blocks(total, 256), 8192)),
+                         dim3(256), 0, cur_stream(), x.data_ptr<scalar_t>(),
+                         mean.data_ptr<float>(), W.data_ptr<float>(),
+                         has_affine ? gamma.data_ptr<scalar_t>() : nullptr,
+                         has_affine ? beta.data_ptr<scalar_t>() : nullptr,
+                         out.data_ptr<scalar_t>(), (int)C, total, relu ? 1 : 0,
+                         has_affine ? 1 : 0);
+    });
+  });
+}
+
+void whiten_bwd_reduce_cl(Tensor x, Tensor dout, Tensor out, Tensor mean,
+                          Tensor W, Tensor gamma, Tensor dWacc, Tensor dgb,
+                          int64_t g, int64_t C, int64_t M, bool relu,
+                          bool has_affine) {
+  const int zslices = (C + 255) / 256;
+  DISPATCH_FT(x, "whiten_bwd_reduce_cl", [&] {
+    DWT_SWITCH_G(g, {
+      const int GW = (C < 256 ? C : 256) / G;
+      dim3 grid(nhwc_reduce_blocks(M, 256 / GW, zslices), 1, zslices);
+      hipLaunchKernelGGL((dwt::whiten_bwd_reduce_nhwc_kernel<scalar_t, G>),
+                         grid, dim3(256), 0, cur_stream(),
+                         x.data_ptr<scalar_t>(), dout.data_ptr<scalar_t>(),
+                         out.data_ptr<scalar_t>(), mean.data_ptr<float>(),
+                         W.data_ptr<float>(),
+                         has_affine ? gamma.data_ptr<scalar_t>() : nullptr,
+                         dWacc.data_ptr<float>(), dgb.data_ptr<float>(), (int)C,
+                         M, relu ? 1 : 0, has_affine ? 1 : 0);
+    });
+  });
+}
+
+void whiten_bwd_apply_cl(Tensor x, Tensor dout, Tensor out, Tensor mean,
+                         Tensor W, Tensor gamma, Tensor S, Tensor corr,
+                         Tensor dx, int64_t g, int64_t C, int64_t M, bool relu,
+                         bool has_affine, bool train_stats) {
+  const int64_t total = M * (C / g);
+  DISPATCH_FT(x, "whiten_bwd_apply_cl", [&] {
+    DWT_SWITCH_G(g, {
+      hipLaunchKernelGGL((dwt::whiten_bwd_apply_nhwc_kernel<scalar_t, G>),
+                         dim3(std::min<int64_t>(elementwise_blocks(total, 256), 8192)),
+                         dim3(256), 0, cur_stream(), x.data_ptr<scalar_t>(),
+                         dout.data_ptr<scalar_t>(), out.data_ptr<scalar_t>(),
+                         mean.data_ptr<float>(), W.data_ptr<float>(),
+                         has_affine ? gamma.data_ptr<scalar_t>() : nullptr,
+                         S.data_ptr<float>(), corr.data_ptr<float>(),
+                         dx.data_ptr<scalar_t>(), (int)C, total, relu ? 1 : 0,
+                         has_affine ? 1 : 0, train_stats ? 1 : 0);
+    });
+  });
+}
+
+void bn_stats_cl(Tensor x, Tensor acc, Tensor mean, Tensor istd, Tensor var_unb,
+                 int64_t C, int64_t M, double eps) {
+  const int zslices = (C + 1023) / 1024;
+  DISPATCH_FT(x, "bn_stats_cl", [&] {
+    const int NCH = (C < 1024 ? C : 1024) / 4;
+    dim3 grid(nhwc_reduce_blocks(M, 256 / NCH > 0 ? 256 / NCH : 1, zslices), 1,
+              zslices);
+    hipLaunchKernelGGL((dwt::bn_stats_nhwc_kernel<scalar_t>), grid, dim3(256),
+                       0, cur_stream(), x.data_ptr<scalar_t>(),
+                       acc.data_ptr<float>(), (int)C, M);
+  });
+  const float unb = M > 1 ? (float)M / (float)(M - 1) : 1.f;
+  hipLaunchKernelGGL(dwt::bn_stats_final_kernel, dim3((C + 255) / 256),
+                     dim3(256), 0, cur_stream(), acc.data_ptr<float>(),
+                     mean.data_ptr<float>(), istd.data_ptr<float>(),
+                     var_unb.data_ptr<float>(), (int)C, 1.0f / (float)M, unb,
+                     (float)eps);
+}
+
+void bn_apply_cl(Tensor x, Tensor mean, Tensor istd, Tensor gamma, Tensor beta,
+                 Tensor out, int64_t C, int64_t M, bool relu, bool has_affine) {
+  const int64_t total4 = M * (C / 4);
+  DISPATCH_FT(x, "bn_apply_cl", [&] {
+    hipLaunchKernelGGL((dwt::bn_apply_nhwc_kernel<scalar_t>),
+                       dim3(std::min<int64_t>(elementwise_blocks(total4, 256), 8192)),
+                       dim3(256), 0, cur_stream(), x.data_ptr<scalar_t>(),
+                       mean.data_ptr<float>(), istd.data_ptr<float>(),
+                       has_affine ? gamma.data_ptr<scalar_t>() : nullptr,
+                       has_affine ? beta.data_ptr<scalar_t>() : nullptr,
+                       out.data_ptr<scalar_t>(), (int)C, total4, relu ? 1 : 0,
+                       has_affine ? 1 : 0);
+  });
+}
+
+void bn_bwd_reduce_cl(Tensor x, Tensor dout, Tensor out, Tensor mean,
+                      Tensor istd, Tensor sums, int64_t C, int64_t M,
+                      bool relu) {
+  const int zslices = (C + 1023) / 1024;
+  DISPATCH_FT(x, "bn_bwd_reduce_cl", [&] {
+    const int NCH = (C < 1024 ? C : 1024) / 4;
+    dim3 grid(nhwc_reduce_blocks(M, 256 / NCH > 0 ? 256 / NCH : 1, zslices), 1,
+              zslices);
+    hipLaunchKernelGGL((dwt::bn_bwd_reduce_nhwc_kernel<scalar_t>), grid,
+                       dim3(256), 0, cur_stream(), x.data_ptr<scalar_t>(),
+                       dout.data_ptr<scalar_t>(), out.data_ptr<scalar_t>(),
+                       mean.data_ptr<float>(), istd.data_ptr<float>(),
+                       sums.data_ptr<float>(), (int)C, M, relu ? 1 : 0);
+  });
+}
+
+void bn_bwd_apply_cl(Tensor x, Tensor dout, Tensor out, Tensor mean,
+                     Tensor istd, Tensor gamma, Tensor sums, Tensor dx,
+                     int64_t C, int64_t M, bool relu, bool has_affine,
+                     bool use_batch) {
+  const int64_t total4 = M * (C / 4);
+  DISPATCH_FT(x, "bn_bwd_apply_cl", [&] {
+    hipLaunchKernelGGL((dwt::bn_bwd_apply_nhwc_kernel<scalar_t>),
+                       dim3(std::min<int64_t>(elementwise_blocks(total4, 256), 8192)),
+                       dim3(256), 0, cur_stream(), x.data_ptr<scalar_t>(),
+                       dout.data_ptr<scalar_t>(), out.data_ptr<scalar_t>(),
+                       mean.data_ptr<float>(), istd.data_ptr<float>(),
+                       has_affine ? gamma.data_ptr<scalar_t>() : nullptr,
+                       sums.data_ptr<float>(), dx.data_ptr<scalar_t>(), (int)C,
+                       total4, 1.0f / (float)M, relu ? 1 : 0,
+                       has_affine ? 1 : 0, use_batch ? 1 : 0);
+  });
+}
+
 void matfn_chol_fwd(Tensor cov, Tensor W, Tensor L, double eps) {
   const int n_groups = cov.size(0);
   const int g = cov.size(1);
@@ -1529,6 +2129,14 @@ void entropy_bwd(Tensor q, Tensor hper, Tensor gscale, Tensor dx) {
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("whiten_stats", &whiten_stats);
+  m.def("whiten_stats_cl", &whiten_stats_cl);
+  m.def("whiten_apply_cl", &whiten_apply_cl);
+  m.def("whiten_bwd_reduce_cl", &whiten_bwd_reduce_cl);
+  m.def("whiten_bwd_apply_cl", &whiten_bwd_apply_cl);
+  m.def("bn_stats_cl", &bn_stats_cl);
+  m.def("bn_apply_cl", &bn_apply_cl);
+  m.def("bn_bwd_reduce_cl", &bn_bwd_reduce_cl);
+  m.def("bn_bwd_apply_cl", &bn_bwd_apply_cl);
   m.def("matfn_chol_fwd", &matfn_chol_fwd);
   m.def("matfn_chol_bwd", &matfn_chol_bwd);
   m.def("matfn_ns_fwd", &matfn_ns_fwd);

@@ -226,3 +226,97 @@ def test_lenet_gpu_step(dev):
     loss.backward()
     opt.step()
     assert torch.isfinite(loss)
+
+
+# ---------------------------------------------------------------------------
+# channels_last (NHWC) kernel variants
+# ---------------------------------------------------------------------------
+
+
+@pytest.mark.parametrize("mode", ["chol", "zca"])
+@pytest.mark.parametrize("c,groups_of", [(64, 16), (256, 64), (512, 128)])
+def test_whiten_channels_last_parity(dev, mode, c, groups_of):
+    torch.manual_seed(10)
+    parts = 3
+    x = torch.randn(parts * 2, c, 6, 10, device=dev)
+    x_cl = x.contiguous(memory_format=torch.channels_last).requires_grad_(True)
+    x_t = x.clone().requires_grad_(True)
+    gamma = torch.randn(c, 1, 1, device=dev)
+    beta = torch.randn(c, 1, 1, device=dev)
+    ga, gb = gamma.clone().requires_grad_(True), gamma.clone().requires_grad_(True)
+    ba, bb = beta.clone().requires_grad_(True), beta.clone().requires_grad_(True)
+    cfg = _whiten_cfg(parts, groups_of, mode=mode, relu=True)
+
+    out_cl = run_whiten(x_cl, ga, ba, cfg, True)
+    out_t = run_whiten(x_t, gb, bb, cfg, False)
+    assert out_cl.is_contiguous(memory_format=torch.channels_last)
+    assert torch.allclose(out_cl, out_t, atol=3e-4), (out_cl - out_t).abs().max()
+
+    gout = torch.randn_like(out_t)
+    out_cl.backward(gout)
+    out_t.backward(gout)
+    assert torch.allclose(x_cl.grad, x_t.grad, atol=2e-3), (x_cl.grad - x_t.grad).abs().max()
+    assert torch.allclose(ga.grad, gb.grad, atol=3e-2)
+    assert torch.allclose(ba.grad, bb.grad, atol=3e-2)
+
+
+def test_whiten_channels_last_bf16(dev):
+    torch.manual_seed(11)
+    c, groups_of = 64, 16
+    x = torch.randn(6, c, 8, 8, device=dev).to(torch.bfloat16)
+    x_cl = x.contiguous(memory_format=torch.channels_last).requires_grad_(True)
+    x_t = x.clone().requires_grad_(True)
+    cfg = _whiten_cfg(3, groups_of)
+    out_cl = run_whiten(x_cl, None, None, cfg, True)
+    out_t = run_whiten(x_t, None, None, cfg, False)
+    assert torch.allclose(out_cl.float(), out_t.float(), atol=0.1, rtol=0.05)
+    gout = torch.randn_like(out_t)
+    out_cl.backward(gout)
+    out_t.backward(gout)
+    assert torch.allclose(x_cl.grad.float(), x_t.grad.float(), atol=0.2, rtol=0.1)
+
+
+def test_bn_channels_last_parity(dev):
+    torch.manual_seed(12)
+    from dwt_amd.kernels.hip_ops import _HipBatchNormMulti
+    from dwt_amd.ops.functional import BatchNormMulti
+    c, parts = 128, 3
+    x = torch.randn(parts * 4, c, 7, 5, device=dev)
+    x_cl = x.contiguous(memory_format=torch.channels_last).requires_grad_(True)
+    x_t = x.clone().requires_grad_(True)
+    gamma = torch.randn(c, 1, 1, device=dev)
+    beta = torch.randn(c, 1, 1, device=dev)
+    ga, gb = gamma.clone().requires_grad_(True), gamma.clone().requires_grad_(True)
+    ba, bb = beta.clone().requires_grad_(True), beta.clone().requires_grad_(True)
+    rms_h = [torch.zeros(c, device=dev) for _ in range(parts)]
+    rvs_h = [torch.ones(c, device=dev) for _ in range(parts)]
+    rms_t = [t.clone() for t in rms_h]
+    rvs_t = [t.clone() for t in rvs_h]
+    cfg = dict(parts=parts, eps=1e-5, momentum=0.1, training=True, relu=True)
+    out_cl = _HipBatchNormMulti.apply(x_cl, ga, ba, rms_h, rvs_h, cfg)
+    out_t = BatchNormMulti.apply(x_t, gb, bb, rms_t, rvs_t, cfg)
+    assert torch.allclose(out_cl, out_t, atol=1e-4)
+    for h, t in zip(rms_h + rvs_h, rms_t + rvs_t):
+        assert torch.allclose(h, t, atol=1e-4)
+    gout = torch.randn_like(out_t)
+    out_cl.backward(gout)
+    out_t.backward(gout)
+    assert torch.allclose(x_cl.grad, x_t.grad, atol=1e-3)
+    assert torch.allclose(ga.grad, gb.grad, atol=1e-2)
+
+
+def test_model_channels_last_step(dev):
+    import torch.nn.functional as F
+    from dwt_amd.models import Bottleneck, ResNetDWT
+    from dwt_amd.ops import functional as Fdwt
+    torch.manual_seed(1)
+    model = ResNetDWT(Bottleneck, [1, 1, 1, 1], None, num_classes=7)
+    model = model.to(dev).to(torch.bfloat16).to(memory_format=torch.channels_last).train()
+    data = torch.randn(6, 3, 64, 64, device=dev, dtype=torch.bfloat16) \
+        .contiguous(memory_format=torch.channels_last)
+    labels = torch.randint(0, 7, (2,), device=dev)
+    out = model(data)
+    s, t, td = torch.split(out, 2, dim=0)
+    loss = F.nll_loss(F.log_softmax(s.float(), 1), labels) + 0.1 * Fdwt.mec_loss(t, td)
+    loss.backward()
+    assert torch.isfinite(loss)
