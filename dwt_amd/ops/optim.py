@@ -62,7 +62,12 @@ class _FusedBase(Optimizer):
             for p in group["params"]:
                 if p.grad is None:
                     continue
-                assert p.grad.is_contiguous() and p.is_contiguous()
+                # the update kernel walks flat dense storage; any dense layout
+                # (incl. channels_last) works as long as param and grad agree
+                dense = p.is_contiguous() or \
+                    (p.dim() == 4 and p.is_contiguous(memory_format=torch.channels_last))
+                assert dense and p.stride() == p.grad.stride(), \
+                    "FusedOptim: param/grad layout mismatch or non-dense param"
                 if bf16 is None:
                     bf16 = p.dtype == torch.bfloat16
                 assert (p.dtype == torch.bfloat16) == bf16, \
@@ -152,7 +157,8 @@ class FusedSGD(_FusedBase):
                     st["momentum_buffer"] = torch.zeros(
                         p.shape, dtype=torch.float32, device=p.device)
                 if p.dtype == torch.bfloat16 and "master" not in st:
-                    st["master"] = p.detach().float().clone()
+                    # same storage order as p (chunk offsets must line up)
+                    st["master"] = torch.empty_like(p, dtype=torch.float32).copy_(p)
 
     def _state_tensors(self, p):
         st = self.state[p]
@@ -204,7 +210,8 @@ class FusedAdam(_FusedBase):
                     st["exp_avg_sq"] = torch.zeros(p.shape, dtype=torch.float32,
                                                    device=p.device)
                 if p.dtype == torch.bfloat16 and "master" not in st:
-                    st["master"] = p.detach().float().clone()
+                    # same storage order as p (chunk offsets must line up)
+                    st["master"] = torch.empty_like(p, dtype=torch.float32).copy_(p)
 
     def _state_tensors(self, p):
         st = self.state[p]
