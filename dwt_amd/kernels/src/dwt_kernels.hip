@@ -872,31 +872,43 @@ __global__ void whiten_stats_nhwc_kernel(
   }
 }
 
-// fused apply, NHWC: lanes tile (position, group) flat
+// fused apply, NHWC: fixed group per thread (W/params in registers),
+// positions advance additively — same CW/GW mapping as the stats kernel.
 template <typename T, int G>
 __global__ void whiten_apply_nhwc_kernel(
     const T* __restrict__ x, const float* __restrict__ mean,
     const float* __restrict__ W, const T* __restrict__ gamma,
     const T* __restrict__ beta, T* __restrict__ out,
-    int C, int64_t total /* = M * n_groups */, int relu, int has_affine) {
-  const int n_groups = C / G;
-  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
-  for (int64_t f = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; f < total;
-       f += stride) {
-    const int grp = (int)(f % n_groups);
-    const int64_t m = f / n_groups;
-    const int c0 = grp * G;
+    int C, int64_t M, int relu, int has_affine) {
+  const int CW = C < 256 ? C : 256;
+  const int GW = CW / G;
+  const int c0 = blockIdx.z * 256 + (threadIdx.x % GW) * G;
+  const int grp = c0 / G;
+  const int rows_per_iter = blockDim.x / GW;
+  const int row_in_block = threadIdx.x / GW;
+
+  float m_[G], Wr[G][G], gm[G], bt[G];
+#pragma unroll
+  for (int i = 0; i < G; ++i) {
+    m_[i] = mean[c0 + i];
+    gm[i] = has_affine ? ldf(gamma + c0 + i) : 1.f;
+    bt[i] = has_affine ? ldf(beta + c0 + i) : 0.f;
+#pragma unroll
+    for (int j = 0; j < G; ++j) Wr[i][j] = W[((int64_t)grp * G + i) * G + j];
+  }
+  const int64_t stride = (int64_t)gridDim.x * rows_per_iter;
+  for (int64_t m = (int64_t)blockIdx.x * rows_per_iter + row_in_block; m < M;
+       m += stride) {
     float v[G], y[G];
     load_group<T, G>(x + m * C + c0, v);
-    const float* Wg = W + (int64_t)grp * G * G;
 #pragma unroll
-    for (int j = 0; j < G; ++j) v[j] -= mean[c0 + j];
+    for (int j = 0; j < G; ++j) v[j] -= m_[j];
 #pragma unroll
     for (int i = 0; i < G; ++i) {
       float a = 0.f;
 #pragma unroll
-      for (int j = 0; j < G; ++j) a += Wg[i * G + j] * v[j];
-      if (has_affine) a = a * ldf(gamma + c0 + i) + ldf(beta + c0 + i);
+      for (int j = 0; j < G; ++j) a += Wr[i][j] * v[j];
+      a = a * gm[i] + bt[i];
       y[i] = relu ? fmaxf(a, 0.f) : a;
     }
     store_group<T, G>(out + m * C + c0, y);
@@ -984,22 +996,37 @@ __global__ void whiten_bwd_reduce_nhwc_kernel(
   }
 }
 
-// backward apply, NHWC
+// backward apply, NHWC (fixed group per thread, W^T/S in registers)
 template <typename T, int G>
 __global__ void whiten_bwd_apply_nhwc_kernel(
     const T* __restrict__ x, const T* __restrict__ dout,
     const T* __restrict__ out, const float* __restrict__ mean,
     const float* __restrict__ W, const T* __restrict__ gamma,
     const float* __restrict__ S, const float* __restrict__ corr,
-    T* __restrict__ dx, int C, int64_t total, int relu, int has_affine,
+    T* __restrict__ dx, int C, int64_t M, int relu, int has_affine,
     int train_stats) {
-  const int n_groups = C / G;
-  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
-  for (int64_t f = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; f < total;
-       f += stride) {
-    const int grp = (int)(f % n_groups);
-    const int64_t m = f / n_groups;
-    const int c0 = grp * G;
+  const int CW = C < 256 ? C : 256;
+  const int GW = CW / G;
+  const int c0 = blockIdx.z * 256 + (threadIdx.x % GW) * G;
+  const int grp = c0 / G;
+  const int rows_per_iter = blockDim.x / GW;
+  const int row_in_block = threadIdx.x / GW;
+
+  float m_[G], Wt[G][G], Sr[G][G], gm[G], cr[G];
+#pragma unroll
+  for (int i = 0; i < G; ++i) {
+    m_[i] = mean[c0 + i];
+    gm[i] = has_affine ? ldf(gamma + c0 + i) : 1.f;
+    cr[i] = train_stats ? corr[c0 + i] : 0.f;
+#pragma unroll
+    for (int j = 0; j < G; ++j) {
+      Wt[i][j] = W[((int64_t)grp * G + j) * G + i];  // W^T
+      Sr[i][j] = train_stats ? S[((int64_t)grp * G + i) * G + j] : 0.f;
+    }
+  }
+  const int64_t stride = (int64_t)gridDim.x * rows_per_iter;
+  for (int64_t m = (int64_t)blockIdx.x * rows_per_iter + row_in_block; m < M;
+       m += stride) {
     float xv[G], dyv[G], ov[G], r[G];
     load_group<T, G>(x + m * C + c0, xv);
     load_group<T, G>(dout + m * C + c0, dyv);
@@ -1008,21 +1035,16 @@ __global__ void whiten_bwd_apply_nhwc_kernel(
 #pragma unroll
       for (int j = 0; j < G; ++j) dyv[j] = ov[j] > 0.f ? dyv[j] : 0.f;
     }
-    const float* Wg = W + (int64_t)grp * G * G;
-    const float* Sg = S + (int64_t)grp * G * G;
 #pragma unroll
     for (int j = 0; j < G; ++j) {
-      xv[j] -= mean[c0 + j];
-      if (has_affine) dyv[j] *= ldf(gamma + c0 + j);
+      xv[j] -= m_[j];
+      dyv[j] *= gm[j];
     }
 #pragma unroll
     for (int i = 0; i < G; ++i) {
-      float a = train_stats ? -corr[c0 + i] : 0.f;
+      float a = -cr[i];
 #pragma unroll
-      for (int j = 0; j < G; ++j) {
-        a += Wg[j * G + i] * dyv[j];
-        if (train_stats) a += Sg[i * G + j] * xv[j];
-      }
+      for (int j = 0; j < G; ++j) a += Wt[i][j] * dyv[j] + Sr[i][j] * xv[j];
       r[i] = a;
     }
     store_group<T, G>(dx + m * C + c0, r);
@@ -1050,10 +1072,20 @@ __global__ void bn_stats_nhwc_kernel(
 #pragma unroll
     for (int k = 0; k < VC; ++k) { s[k] += v[k]; ss[k] += v[k] * v[k]; }
   }
+  // per-block LDS accumulation -> one global atomic per channel per block
+  __shared__ float lacc[2 * 1024];
+  const int cb = c0 - blockIdx.z * 1024;
+  for (int k = threadIdx.x; k < 2 * CW; k += blockDim.x) lacc[k] = 0.f;
+  __syncthreads();
 #pragma unroll
   for (int k = 0; k < VC; ++k) {
-    atomicAdd(&acc[c0 + k], s[k]);
-    atomicAdd(&acc[C + c0 + k], ss[k]);
+    atomicAdd(&lacc[cb + k], s[k]);
+    atomicAdd(&lacc[CW + cb + k], ss[k]);
+  }
+  __syncthreads();
+  for (int k = threadIdx.x; k < CW; k += blockDim.x) {
+    atomicAdd(&acc[blockIdx.z * 1024 + k], lacc[k]);
+    atomicAdd(&acc[C + blockIdx.z * 1024 + k], lacc[CW + k]);
   }
 }
 
@@ -1061,21 +1093,30 @@ template <typename T>
 __global__ void bn_apply_nhwc_kernel(
     const T* __restrict__ x, const float* __restrict__ mean,
     const float* __restrict__ istd, const T* __restrict__ gamma,
-    const T* __restrict__ beta, T* __restrict__ out, int C, int64_t total4,
+    const T* __restrict__ beta, T* __restrict__ out, int C, int64_t M,
     int relu, int has_affine) {
   constexpr int VC = 4;
-  const int nch = C / VC;
-  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
-  for (int64_t f = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; f < total4;
-       f += stride) {
-    const int c0 = (int)(f % nch) * VC;
-    const int64_t m = f / nch;
+  const int CW = C < 1024 ? C : 1024;
+  const int NCH = CW / VC;
+  const int c0 = blockIdx.z * 1024 + (threadIdx.x % NCH) * VC;
+  const int rows_per_iter = blockDim.x / NCH;
+  const int row_in_block = threadIdx.x / NCH;
+  float mu[VC], is[VC], gm[VC], bt[VC];
+#pragma unroll
+  for (int k = 0; k < VC; ++k) {
+    mu[k] = mean[c0 + k];
+    is[k] = istd[c0 + k];
+    gm[k] = has_affine ? ldf(gamma + c0 + k) : 1.f;
+    bt[k] = has_affine ? ldf(beta + c0 + k) : 0.f;
+  }
+  const int64_t stride = (int64_t)gridDim.x * rows_per_iter;
+  for (int64_t m = (int64_t)blockIdx.x * rows_per_iter + row_in_block; m < M;
+       m += stride) {
     float v[VC];
     load_group<T, VC>(x + m * C + c0, v);
 #pragma unroll
     for (int k = 0; k < VC; ++k) {
-      float y = (v[k] - mean[c0 + k]) * istd[c0 + k];
-      if (has_affine) y = y * ldf(gamma + c0 + k) + ldf(beta + c0 + k);
+      const float y = (v[k] - mu[k]) * is[k] * gm[k] + bt[k];
       v[k] = relu ? fmaxf(y, 0.f) : y;
     }
     store_group<T, VC>(out + m * C + c0, v);
@@ -1114,10 +1155,19 @@ __global__ void bn_bwd_reduce_nhwc_kernel(
       s_dyxh[k] += dv[k] * (xv[k] - mean[c0 + k]) * istd[c0 + k];
     }
   }
+  __shared__ float lacc[2 * 1024];
+  const int cb = c0 - blockIdx.z * 1024;
+  for (int k = threadIdx.x; k < 2 * CW; k += blockDim.x) lacc[k] = 0.f;
+  __syncthreads();
 #pragma unroll
   for (int k = 0; k < VC; ++k) {
-    atomicAdd(&sums[c0 + k], s_dy[k]);
-    atomicAdd(&sums[C + c0 + k], s_dyxh[k]);
+    atomicAdd(&lacc[cb + k], s_dy[k]);
+    atomicAdd(&lacc[CW + cb + k], s_dyxh[k]);
+  }
+  __syncthreads();
+  for (int k = threadIdx.x; k < CW; k += blockDim.x) {
+    atomicAdd(&sums[blockIdx.z * 1024 + k], lacc[k]);
+    atomicAdd(&sums[C + blockIdx.z * 1024 + k], lacc[CW + k]);
   }
 }
 
@@ -1126,15 +1176,27 @@ __global__ void bn_bwd_apply_nhwc_kernel(
     const T* __restrict__ x, const T* __restrict__ dout,
     const T* __restrict__ out, const float* __restrict__ mean,
     const float* __restrict__ istd, const T* __restrict__ gamma,
-    const float* __restrict__ sums, T* __restrict__ dx, int C, int64_t total4,
+    const float* __restrict__ sums, T* __restrict__ dx, int C, int64_t M,
     float inv_m, int relu, int has_affine, int use_batch) {
   constexpr int VC = 4;
-  const int nch = C / VC;
-  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
-  for (int64_t f = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; f < total4;
-       f += stride) {
-    const int c0 = (int)(f % nch) * VC;
-    const int64_t m = f / nch;
+  const int CW = C < 1024 ? C : 1024;
+  const int NCH = CW / VC;
+  const int c0 = blockIdx.z * 1024 + (threadIdx.x % NCH) * VC;
+  const int rows_per_iter = blockDim.x / NCH;
+  const int row_in_block = threadIdx.x / NCH;
+  float mu[VC], is[VC], gm[VC], mdy[VC], mdyxh[VC];
+#pragma unroll
+  for (int k = 0; k < VC; ++k) {
+    const int c = c0 + k;
+    mu[k] = mean[c];
+    is[k] = istd[c];
+    gm[k] = has_affine ? ldf(gamma + c) : 1.f;
+    mdy[k] = use_batch ? sums[c] * inv_m : 0.f;
+    mdyxh[k] = use_batch ? sums[C + c] * inv_m : 0.f;
+  }
+  const int64_t stride = (int64_t)gridDim.x * rows_per_iter;
+  for (int64_t m = (int64_t)blockIdx.x * rows_per_iter + row_in_block; m < M;
+       m += stride) {
     float xv[VC], dv[VC], ov[VC];
     load_group<T, VC>(x + m * C + c0, xv);
     load_group<T, VC>(dout + m * C + c0, dv);
@@ -1145,15 +1207,12 @@ __global__ void bn_bwd_apply_nhwc_kernel(
     }
 #pragma unroll
     for (int k = 0; k < VC; ++k) {
-      const int c = c0 + k;
-      const float gm = has_affine ? ldf(gamma + c) : 1.f;
-      const float is = istd[c];
-      const float dxh = dv[k] * gm;
+      const float dxh = dv[k] * gm[k];
       if (use_batch) {
-        const float xh = (xv[k] - mean[c]) * is;
-        dv[k] = (dxh - gm * sums[c] * inv_m - xh * gm * sums[C + c] * inv_m) * is;
+        const float xh = (xv[k] - mu[k]) * is[k];
+        dv[k] = (dxh - gm[k] * mdy[k] - xh * gm[k] * mdyxh[k]) * is[k];
       } else {
-        dv[k] = dxh * is;
+        dv[k] = dxh * is[k];
       }
     }
     store_group<T, VC>(dx + m * C + c0, dv);
@@ -1646,19 +1705,26 @@ void whiten_stats_cl(Tensor x, Tensor acc, Tensor mean, Tensor cov, int64_t g,
                      (int)g, 1.0f / (float)M);
 }
 
+inline int64_t nhwc_elem_blocks(int64_t M, int rows_per_iter, int zslices) {
+  const int64_t want = (M + rows_per_iter - 1) / rows_per_iter;
+  const int64_t cap = std::max<int64_t>(1, 8192 / std::max(zslices, 1));
+  return std::min(std::max<int64_t>(want, 1), cap);
+}
+
 void whiten_apply_cl(Tensor x, Tensor mean, Tensor W, Tensor gamma, Tensor beta,
                      Tensor out, int64_t g, int64_t C, int64_t M, bool relu,
                      bool has_affine) {
-  const int64_t total = M * (C / g);
+  const int zslices = (C + 255) / 256;
   DISPATCH_FT(x, "whiten_apply_cl", [&] {
     DWT_SWITCH_G(g, {
-      hipLaunchKernelGGL((dwt::whiten_apply_nhwc_kernel<scalar_t, G>),
-                         dim3(std::min<int64_t>(elementwise_blocks(total, 256), 8192)),
+      const int GW = (C < 256 ? C : 256) / G;
+      dim3 grid(nhwc_elem_blocks(M, 256 / GW, zslices), 1, zslices);
+      hipLaunchKernelGGL((dwt::whiten_apply_nhwc_kernel<scalar_t, G>), grid,
                          dim3(256), 0, cur_stream(), x.data_ptr<scalar_t>(),
                          mean.data_ptr<float>(), W.data_ptr<float>(),
                          has_affine ? gamma.data_ptr<scalar_t>() : nullptr,
                          has_affine ? beta.data_ptr<scalar_t>() : nullptr,
-                         out.data_ptr<scalar_t>(), (int)C, total, relu ? 1 : 0,
+                         out.data_ptr<scalar_t>(), (int)C, M, relu ? 1 : 0,
                          has_affine ? 1 : 0);
     });
   });
@@ -1689,17 +1755,19 @@ void whiten_bwd_apply_cl(Tensor x, Tensor dout, Tensor out, Tensor mean,
                          Tensor W, Tensor gamma, Tensor S, Tensor corr,
                          Tensor dx, int64_t g, int64_t C, int64_t M, bool relu,
                          bool has_affine, bool train_stats) {
-  const int64_t total = M * (C / g);
+  const int zslices = (C + 255) / 256;
   DISPATCH_FT(x, "whiten_bwd_apply_cl", [&] {
     DWT_SWITCH_G(g, {
+      const int GW = (C < 256 ? C : 256) / G;
+      dim3 grid(nhwc_elem_blocks(M, 256 / GW, zslices), 1, zslices);
       hipLaunchKernelGGL((dwt::whiten_bwd_apply_nhwc_kernel<scalar_t, G>),
-                         dim3(std::min<int64_t>(elementwise_blocks(total, 256), 8192)),
-                         dim3(256), 0, cur_stream(), x.data_ptr<scalar_t>(),
+                         grid, dim3(256), 0, cur_stream(),
+                         x.data_ptr<scalar_t>(),
                          dout.data_ptr<scalar_t>(), out.data_ptr<scalar_t>(),
                          mean.data_ptr<float>(), W.data_ptr<float>(),
                          has_affine ? gamma.data_ptr<scalar_t>() : nullptr,
                          S.data_ptr<float>(), corr.data_ptr<float>(),
-                         dx.data_ptr<scalar_t>(), (int)C, total, relu ? 1 : 0,
+                         dx.data_ptr<scalar_t>(), (int)C, M, relu ? 1 : 0,
                          has_affine ? 1 : 0, train_stats ? 1 : 0);
     });
   });
@@ -1726,15 +1794,17 @@ void bn_stats_cl(Tensor x, Tensor acc, Tensor mean, Tensor istd, Tensor var_unb,
 
 void bn_apply_cl(Tensor x, Tensor mean, Tensor istd, Tensor gamma, Tensor beta,
                  Tensor out, int64_t C, int64_t M, bool relu, bool has_affine) {
-  const int64_t total4 = M * (C / 4);
+  const int zslices = (C + 1023) / 1024;
   DISPATCH_FT(x, "bn_apply_cl", [&] {
-    hipLaunchKernelGGL((dwt::bn_apply_nhwc_kernel<scalar_t>),
-                       dim3(std::min<int64_t>(elementwise_blocks(total4, 256), 8192)),
+    const int NCH = (C < 1024 ? C : 1024) / 4;
+    const int rpi = std::max(256 / NCH, 1);
+    dim3 grid(nhwc_elem_blocks(M, rpi, zslices), 1, zslices);
+    hipLaunchKernelGGL((dwt::bn_apply_nhwc_kernel<scalar_t>), grid,
                        dim3(256), 0, cur_stream(), x.data_ptr<scalar_t>(),
                        mean.data_ptr<float>(), istd.data_ptr<float>(),
                        has_affine ? gamma.data_ptr<scalar_t>() : nullptr,
                        has_affine ? beta.data_ptr<scalar_t>() : nullptr,
-                       out.data_ptr<scalar_t>(), (int)C, total4, relu ? 1 : 0,
+                       out.data_ptr<scalar_t>(), (int)C, M, relu ? 1 : 0,
                        has_affine ? 1 : 0);
   });
 }
@@ -1759,16 +1829,18 @@ void bn_bwd_apply_cl(Tensor x, Tensor dout, Tensor out, Tensor mean,
                      Tensor istd, Tensor gamma, Tensor sums, Tensor dx,
                      int64_t C, int64_t M, bool relu, bool has_affine,
                      bool use_batch) {
-  const int64_t total4 = M * (C / 4);
+  const int zslices = (C + 1023) / 1024;
   DISPATCH_FT(x, "bn_bwd_apply_cl", [&] {
-    hipLaunchKernelGGL((dwt::bn_bwd_apply_nhwc_kernel<scalar_t>),
-                       dim3(std::min<int64_t>(elementwise_blocks(total4, 256), 8192)),
+    const int NCH = (C < 1024 ? C : 1024) / 4;
+    const int rpi = std::max(256 / NCH, 1);
+    dim3 grid(nhwc_elem_blocks(M, rpi, zslices), 1, zslices);
+    hipLaunchKernelGGL((dwt::bn_bwd_apply_nhwc_kernel<scalar_t>), grid,
                        dim3(256), 0, cur_stream(), x.data_ptr<scalar_t>(),
                        dout.data_ptr<scalar_t>(), out.data_ptr<scalar_t>(),
                        mean.data_ptr<float>(), istd.data_ptr<float>(),
                        has_affine ? gamma.data_ptr<scalar_t>() : nullptr,
                        sums.data_ptr<float>(), dx.data_ptr<scalar_t>(), (int)C,
-                       total4, 1.0f / (float)M, relu ? 1 : 0,
+                       M, 1.0f / (float)M, relu ? 1 : 0,
                        has_affine ? 1 : 0, use_batch ? 1 : 0);
   });
 }
