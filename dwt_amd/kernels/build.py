@@ -19,7 +19,8 @@ ARCH = os.environ.get("PYTORCH_ROCM_ARCH", "gfx950")
 
 
 def build(verbose: bool = True) -> str | None:
-    sources = sorted(glob.glob(os.path.join(SRC, "*.hip"))) + \
+    sources = sorted(s for s in glob.glob(os.path.join(SRC, "*.hip"))
+                     if not s.endswith("_hip.hip")) + \
         sorted(glob.glob(os.path.join(SRC, "*.cpp")))
     if not sources:
         print("dwt_amd.kernels.build: no HIP sources yet; nothing to build")

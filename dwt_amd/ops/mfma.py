@@ -1,0 +1,108 @@
+"""Python layer over the hand-written MFMA GEMM / implicit-GEMM conv kernels.
+
+``mfma_gemm``: C[M,N] = A[M,K] @ B[N,K]^T (+bias/+ReLU), bf16 in, fp32
+accumulate, bf16 out — the v_mfma_f32_16x16x32_bf16 tile kernel.
+
+``MFMALinear``: nn.Linear whose forward AND backward run on mfma_gemm
+(dgrad/wgrad are plain GEMMs with small host-side transposes).
+
+``conv2d_fwd``: NHWC implicit-GEMM convolution forward on the same kernel
+(1x1 degenerates to the dense-GEMM fast path; KxK gathers patches on the
+fly; the channels_last conv weight is consumed as stored).
+"""
+from __future__ import annotations
+
+from typing import Optional
+
+import torch
+import torch.nn as nn
+import torch.nn.functional as F
+
+
+def _ext():
+    from ..kernels import dispatch
+    return dispatch.ext()
+
+
+def _pad_k(t: torch.Tensor) -> torch.Tensor:
+    k = t.shape[1]
+    if k % 8 == 0:
+        return t
+    return F.pad(t, (0, 8 - k % 8))
+
+
+def mfma_gemm(a: torch.Tensor, bt: torch.Tensor,
+              bias: Optional[torch.Tensor] = None, relu: bool = False):
+    """a (M,K) bf16, bt (N,K) bf16 -> (M,N) bf16 = a @ bt.T."""
+    assert a.dtype == torch.bfloat16 and bt.dtype == torch.bfloat16
+    a = _pad_k(a.contiguous())
+    bt = _pad_k(bt.contiguous())
+    m, n = a.shape[0], bt.shape[0]
+    out = torch.empty(m, n, device=a.device, dtype=torch.bfloat16)
+    has_bias = bias is not None
+    b32 = bias.detach().float().contiguous() if has_bias else torch.empty(0, device=a.device)
+    _ext().mfma_gemm(a, bt, b32, out, relu, has_bias)
+    return out
+
+
+class _MFMALinearFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, weight, bias):
+        x2 = x.reshape(-1, x.shape[-1])
+        out = mfma_gemm(x2, weight, bias=bias, relu=False)
+        ctx.save_for_backward(x2, weight)
+        ctx.has_bias = bias is not None
+        ctx.in_shape = x.shape
+        return out.reshape(*x.shape[:-1], weight.shape[0])
+
+    @staticmethod
+    def backward(ctx, dout):
+        x2, weight = ctx.saved_tensors
+        dout2 = dout.reshape(-1, dout.shape[-1]).contiguous()
+        # dx = dy @ W          : A = dy (M,N), Bt = W^T (K,N)
+        dx = mfma_gemm(dout2, weight.t().contiguous())
+        # dW = dy^T @ x        : A = dy^T (N,M), Bt = x^T (K,M)
+        dw = mfma_gemm(dout2.t().contiguous(), x2.t().contiguous())
+        db = dout2.sum(0).to(dout.dtype) if ctx.has_bias else None
+        return dx.reshape(ctx.in_shape), dw, db
+
+
+class MFMALinear(nn.Linear):
+    """nn.Linear running fwd/dgrad/wgrad on the hand-written MFMA GEMM when
+    on GPU in bf16; falls back to F.linear otherwise."""
+
+    def forward(self, x):
+        if x.is_cuda and x.dtype == torch.bfloat16 \
+                and self.weight.dtype == torch.bfloat16:
+            from ..kernels import dispatch
+            if dispatch.available():
+                return _MFMALinearFn.apply(x, self.weight, self.bias)
+        return F.linear(x, self.weight, self.bias)
+
+
+def conv2d_fwd(x: torch.Tensor, weight: torch.Tensor,
+               bias: Optional[torch.Tensor] = None, stride: int = 1,
+               padding: int = 0, relu: bool = False) -> torch.Tensor:
+    """NHWC implicit-GEMM conv forward (inference path).
+
+    x: (N, C, H, W) channels_last bf16; weight: (Cout, Cin, KH, KW)
+    channels_last bf16; returns (N, Cout, P, Q) channels_last bf16.
+    """
+    assert x.dtype == torch.bfloat16
+    assert x.is_contiguous(memory_format=torch.channels_last) or x.shape[1] == 1
+    n, cin, h, w = x.shape
+    cout, _, kh, kw = weight.shape
+    p = (h + 2 * padding - kh) // stride + 1
+    q = (w + 2 * padding - kw) // stride + 1
+    out = torch.empty(n, cout, p, q, device=x.device, dtype=torch.bfloat16,
+                      memory_format=torch.channels_last)
+    wgt = weight
+    if not wgt.is_contiguous(memory_format=torch.channels_last):
+        wgt = wgt.contiguous(memory_format=torch.channels_last)
+    has_bias = bias is not None
+    b32 = bias.detach().float().contiguous() if has_bias else torch.empty(0, device=x.device)
+    if (kh * kw * cin) % 8 != 0 and not (kh == 1 and kw == 1 and stride == 1 and padding == 0):
+        pass  # slow per-element gather path inside the kernel handles it
+    _ext().mfma_conv2d_fwd(x, wgt, b32, out, n, h, w, cin, p, q, cout,
+                           kh, kw, stride, padding, relu, has_bias)
+    return out
