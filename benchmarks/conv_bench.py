@@ -5,7 +5,11 @@ MIOpen/rocBLAS on the R50 forward shapes. Writes a markdown table.
 Run on the GPU box:  python benchmarks/conv_bench.py [--out profiles/conv_bench.md]
 """
 import argparse
+import os
+import sys
 import time
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 
 import torch
 import torch.nn.functional as F
