@@ -35,7 +35,8 @@ def train_infinite_collect_stats(args, model, device, source_train_loader,
                                  target_train_loader, optimizer, lambda_mec_loss,
                                  target_test_loader, logger: JsonlLogger = None,
                                  grad_sync=None, start_iter=0,
-                                 checkpoint_path=None, checkpoint_every=0):
+                                 checkpoint_path=None, checkpoint_every=0,
+                                 stats_passes=10):
     source_iter = iter(source_train_loader)
     target_iter = iter(target_train_loader)
     exp_lr_scheduler = lr_scheduler.MultiStepLR(optimizer, milestones=[6000], gamma=0.1)
@@ -92,7 +93,8 @@ def train_infinite_collect_stats(args, model, device, source_train_loader,
 
     print("Training is complete...")
     print("Running a bunch of forward passes to estimate the population statistics of target...")
-    eval_pass_collect_stats(args, model, device, target_test_loader)
+    eval_pass_collect_stats(args, model, device, target_test_loader,
+                            passes=stats_passes)
     print("Finally computing the precision on the test set...")
     return test(args, model, device, target_test_loader, logger=logger)
 

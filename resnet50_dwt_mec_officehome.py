@@ -61,6 +61,8 @@ def build_args(argv=None):
     parser.add_argument('--checkpoint_every', type=int, default=1000)
     parser.add_argument('--resume', action='store_true')
     parser.add_argument('--metrics_jsonl', type=str, default='')
+    parser.add_argument('--stats_passes', type=int, default=10,
+                        help='target-stats re-estimation passes before final test')
     return parser.parse_args(argv)
 
 
@@ -154,7 +156,7 @@ def main(argv=None):
         target_test_loader=test_loader, logger=logger,
         grad_sync=ddp.sync if ddp.enabled else None, start_iter=start_iter,
         checkpoint_path=args.checkpoint_path or None,
-        checkpoint_every=args.checkpoint_every)
+        checkpoint_every=args.checkpoint_every, stats_passes=args.stats_passes)
     logger.close()
 
 
