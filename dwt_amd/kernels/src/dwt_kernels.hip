@@ -1405,6 +1405,50 @@ __global__ void ema_update_kernel(float* __restrict__ rm,
 }
 
 // ===========================================================================
+// Fused residual join: out = relu(a + b); backward: da = db = dout*(out>0)
+// (the reference's `relu(out + identity)` at every bottleneck exit,
+// resnet50_dwt_mec_officehome.py:239-240)
+// ===========================================================================
+
+template <typename T, bool VECTOR>
+__global__ void add_relu_fwd_kernel(const T* __restrict__ a,
+                                    const T* __restrict__ b,
+                                    T* __restrict__ out, int64_t n) {
+  constexpr int VW = VECTOR ? VecTraits<T>::W : 1;
+  const int64_t i = ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) * VW;
+  if (i >= n) return;
+  if (VECTOR) {
+    float va[VW], vb[VW];
+    VecTraits<T>::load(a + i, va);
+    VecTraits<T>::load(b + i, vb);
+#pragma unroll
+    for (int k = 0; k < VW; ++k) va[k] = fmaxf(va[k] + vb[k], 0.f);
+    VecTraits<T>::store(out + i, va);
+  } else {
+    stf(out + i, fmaxf(ldf(a + i) + ldf(b + i), 0.f));
+  }
+}
+
+template <typename T, bool VECTOR>
+__global__ void add_relu_bwd_kernel(const T* __restrict__ dout,
+                                    const T* __restrict__ out,
+                                    T* __restrict__ din, int64_t n) {
+  constexpr int VW = VECTOR ? VecTraits<T>::W : 1;
+  const int64_t i = ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) * VW;
+  if (i >= n) return;
+  if (VECTOR) {
+    float vd[VW], vo[VW];
+    VecTraits<T>::load(dout + i, vd);
+    VecTraits<T>::load(out + i, vo);
+#pragma unroll
+    for (int k = 0; k < VW; ++k) vd[k] = vo[k] > 0.f ? vd[k] : 0.f;
+    VecTraits<T>::store(din + i, vd);
+  } else {
+    stf(din + i, ldf(out + i) > 0.f ? ldf(dout + i) : 0.f);
+  }
+}
+
+// ===========================================================================
 // Fused multi-tensor optimizers.
 // One launch per param-group: grid.x indexes chunk descriptors
 // [p_ptr, g_ptr, state1_ptr, state2_ptr(or master), n].  bf16 params keep an
@@ -2106,6 +2150,45 @@ void bn_bwd_apply(Tensor x, Tensor dout, Tensor out, Tensor mean, Tensor istd,
   });
 }
 
+void add_relu_fwd(Tensor a, Tensor b, Tensor out) {
+  const int64_t n = a.numel();
+  DISPATCH_FT(a, "add_relu_fwd", [&] {
+    constexpr int VW = dwt::VecTraits<scalar_t>::W;
+    const bool vec = (n % VW == 0) &&
+        (reinterpret_cast<uintptr_t>(a.data_ptr()) % 16 == 0) &&
+        (reinterpret_cast<uintptr_t>(b.data_ptr()) % 16 == 0);
+    const int64_t per = vec ? (n + 256 * VW - 1) / (256 * VW)
+                            : (n + 255) / 256;
+    if (vec)
+      hipLaunchKernelGGL((dwt::add_relu_fwd_kernel<scalar_t, true>), dim3(per),
+                         dim3(256), 0, cur_stream(), a.data_ptr<scalar_t>(),
+                         b.data_ptr<scalar_t>(), out.data_ptr<scalar_t>(), n);
+    else
+      hipLaunchKernelGGL((dwt::add_relu_fwd_kernel<scalar_t, false>), dim3(per),
+                         dim3(256), 0, cur_stream(), a.data_ptr<scalar_t>(),
+                         b.data_ptr<scalar_t>(), out.data_ptr<scalar_t>(), n);
+  });
+}
+
+void add_relu_bwd(Tensor dout, Tensor out, Tensor din) {
+  const int64_t n = dout.numel();
+  DISPATCH_FT(dout, "add_relu_bwd", [&] {
+    constexpr int VW = dwt::VecTraits<scalar_t>::W;
+    const bool vec = (n % VW == 0) &&
+        (reinterpret_cast<uintptr_t>(dout.data_ptr()) % 16 == 0);
+    const int64_t per = vec ? (n + 256 * VW - 1) / (256 * VW)
+                            : (n + 255) / 256;
+    if (vec)
+      hipLaunchKernelGGL((dwt::add_relu_bwd_kernel<scalar_t, true>), dim3(per),
+                         dim3(256), 0, cur_stream(), dout.data_ptr<scalar_t>(),
+                         out.data_ptr<scalar_t>(), din.data_ptr<scalar_t>(), n);
+    else
+      hipLaunchKernelGGL((dwt::add_relu_bwd_kernel<scalar_t, false>), dim3(per),
+                         dim3(256), 0, cur_stream(), dout.data_ptr<scalar_t>(),
+                         out.data_ptr<scalar_t>(), din.data_ptr<scalar_t>(), n);
+  });
+}
+
 void fused_sgd(Tensor desc, int64_t n_chunks, double lr, double momentum,
                double wd, bool bf16_params, bool master, bool zero_grad,
                bool first_step) {
@@ -2232,6 +2315,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("bn_bwd_reduce", &bn_bwd_reduce);
   m.def("bn_bwd_apply", &bn_bwd_apply);
   m.def("ema_update", &ema_update);
+  m.def("add_relu_fwd", &add_relu_fwd);
+  m.def("add_relu_bwd", &add_relu_bwd);
   m.def("fused_sgd", &fused_sgd);
   m.def("fused_adam", &fused_adam);
   m.def("mec_fwd", &mec_fwd);

@@ -25,6 +25,7 @@ import torch.nn as nn
 
 from ..ops.batch_norm import DomainBatchNorm2d
 from ..ops.whitening import WhiteningScaleShift
+from ..ops.functional import add_relu
 from .sites import norm_site
 
 
@@ -134,7 +135,7 @@ class Bottleneck(nn.Module):
                 [self.downsample_bns, self.downsample_bnt, self.downsample_bnt_aug],
                 self.downsample_gamma, self.downsample_beta, training=tr, relu=False)
 
-        return torch.relu(out + identity)
+        return add_relu(out, identity)
 
 
 class ResNetDWT(nn.Module):

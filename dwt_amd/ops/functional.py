@@ -501,3 +501,59 @@ def entropy_loss(x: torch.Tensor) -> torch.Tensor:
         if dispatch.available():
             return dispatch.entropy_loss(x)
     return EntropyLossFn.apply(x)
+
+
+# ----------------------------------------------------------------------------
+# Fused residual join: relu(a + b)
+# ----------------------------------------------------------------------------
+
+
+class AddReluFn(torch.autograd.Function):
+    """out = relu(a + b); da = db = dout * (out > 0) — one kernel each way on
+    GPU (the reference's bottleneck exit, resnet50_dwt_mec_officehome.py:239)."""
+
+    @staticmethod
+    def _mf(t):
+        if t.dim() == 4 and t.is_contiguous(memory_format=torch.channels_last) \
+                and not t.is_contiguous():
+            return torch.channels_last
+        return torch.contiguous_format
+
+    @staticmethod
+    def forward(ctx, a, b):
+        if a.is_cuda and a.dtype in (torch.float32, torch.bfloat16):
+            from ..kernels import dispatch
+            if dispatch.available():
+                ext = dispatch.ext()
+                mf = AddReluFn._mf(a)
+                ac = a.contiguous(memory_format=mf) if a.dim() == 4 else a.contiguous()
+                bc = b.contiguous(memory_format=mf) if b.dim() == 4 else b.contiguous()
+                out = torch.empty_like(ac)
+                ext.add_relu_fwd(ac, bc, out)
+                ctx.save_for_backward(out)
+                ctx.use_hip = True
+                return out
+        out = torch.relu(a + b)
+        ctx.save_for_backward(out)
+        ctx.use_hip = False
+        return out
+
+    @staticmethod
+    def backward(ctx, dout):
+        (out,) = ctx.saved_tensors
+        if ctx.use_hip:
+            from ..kernels import dispatch
+            ext = dispatch.ext()
+            mf = AddReluFn._mf(out)
+            dout = dout.contiguous(memory_format=mf) if dout.dim() == 4 \
+                else dout.contiguous()
+            din = torch.empty_like(out)
+            ext.add_relu_bwd(dout, out, din)
+            return din, din
+        mask = (out > 0).to(dout.dtype)
+        d = dout * mask
+        return d, d
+
+
+def add_relu(a: torch.Tensor, b: torch.Tensor) -> torch.Tensor:
+    return AddReluFn.apply(a, b)
