@@ -320,3 +320,24 @@ def test_model_channels_last_step(dev):
     loss = F.nll_loss(F.log_softmax(s.float(), 1), labels) + 0.1 * Fdwt.mec_loss(t, td)
     loss.backward()
     assert torch.isfinite(loss)
+
+
+def test_add_relu_parity(dev):
+    from dwt_amd.ops.functional import AddReluFn
+    torch.manual_seed(13)
+    for mf in (torch.contiguous_format, torch.channels_last):
+        a = torch.randn(4, 64, 8, 8, device=dev, dtype=torch.bfloat16,
+                        requires_grad=True)
+        b = torch.randn(4, 64, 8, 8, device=dev, dtype=torch.bfloat16,
+                        requires_grad=True)
+        a2 = a.detach().clone().requires_grad_(True)
+        b2 = b.detach().clone().requires_grad_(True)
+        av = a.contiguous(memory_format=mf)
+        out = AddReluFn.apply(av, b.contiguous(memory_format=mf))
+        ref = torch.relu(a2 + b2)
+        assert torch.allclose(out.float(), ref.float(), atol=1e-2)
+        g = torch.randn_like(ref)
+        out.backward(g.contiguous(memory_format=mf))
+        ref.backward(g)
+        assert torch.allclose(a.grad.float(), a2.grad.float(), atol=1e-2)
+        assert torch.allclose(b.grad.float(), b2.grad.float(), atol=1e-2)
