@@ -40,6 +40,9 @@ def get_args():
     p.add_argument("--layers", default="3,4,6,3")
     p.add_argument("--channels_last", type=int, default=1,
                    help="NHWC layout (MIOpen-native on gfx950; 0 = NCHW)")
+    p.add_argument("--graph", type=int, default=0,
+                   help="capture the train step in a hipGraph and replay "
+                        "(single-GPU only)")
     return p.parse_args()
 
 
@@ -110,12 +113,26 @@ def main():
         if use_cuda:
             torch.cuda.synchronize()
 
+    use_graph = bool(args.graph) and use_cuda and world == 1
+    if use_graph:
+        # warm up eager (also builds optimizer tables / EMA state), then
+        # capture one steady-state step and replay it
+        for _ in range(max(args.warmup, 2)):
+            step()
+        torch.cuda.synchronize()
+        g = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(g):
+            step()
+        run_step = g.replay
+    else:
+        run_step = step
+
     for _ in range(args.warmup):
-        step()
+        run_step()
     barrier_sync()
     t0 = time.perf_counter()
     for _ in range(args.steps):
-        step()
+        run_step()
     barrier_sync()
     elapsed = time.perf_counter() - t0
 
