@@ -28,7 +28,7 @@ def get_args():
     p.add_argument("--gpus", type=int, default=1)
     p.add_argument("--steps", type=int, default=20)
     p.add_argument("--warmup", type=int, default=5)
-    p.add_argument("--batch", type=int, default=256,
+    p.add_argument("--batch", type=int, default=512,
                    help="per-domain per-GPU batch (total 3x this per GPU; "
                         "288 GB HBM3E comfortably fits 3x256 @224^2 bf16)")
     p.add_argument("--img", type=int, default=224)
