@@ -63,12 +63,18 @@ DEV_INLINE unsigned short f_to_bf16(float f) {
   return (unsigned short)((v.i + 0x7fffu + lsb) >> 16);
 }
 
-// Stage one 128x64 tile into LDS (bf16), rows gathered by a row->global
-// functor; each thread moves 32 halves as 4 chunks of 8.
-// chunk-of-8 stays within one (r,s) patch element when Cin % 8 == 0.
+// Tile staging, split into LOAD (issue global reads into registers, early)
+// and WRITE (LDS store, late) so HBM latency hides under the MFMA phase
+// (guide §6 G15 async-STAGE split / T14).  Each thread moves 32 halves as 4
+// chunks of 8; a chunk-of-8 stays within one (r,s) patch element when
+// Cin % 8 == 0.
+struct StageRegs {
+  uint4 c[4];
+};
+
 template <bool GEMM_FAST>  // true: A is a plain dense [M][K] matrix
-DEV_INLINE void stage_a(const bf16* __restrict__ x, const ConvParams cp,
-                        int64_t m0, int k0, unsigned short* lds) {
+DEV_INLINE void load_a(const bf16* __restrict__ x, const ConvParams& cp,
+                       int64_t m0, int k0, StageRegs& rg) {
   const int t = threadIdx.x;
 #pragma unroll
   for (int c = 0; c < 4; ++c) {
@@ -76,19 +82,15 @@ DEV_INLINE void stage_a(const bf16* __restrict__ x, const ConvParams cp,
     const int row = idx / (BK / 8);           // 0..127
     const int kc = (idx % (BK / 8)) * 8;      // chunk k offset in tile
     const int64_t m = m0 + row;
-    unsigned short* dst = lds + row * LDS_PITCH + kc;
     const int kg = k0 + kc;
     if (m >= cp.M || kg >= cp.K) {
-#pragma unroll
-      for (int j = 0; j < 8; ++j) dst[j] = 0;
+      rg.c[c] = make_uint4(0, 0, 0, 0);
       continue;
     }
     if (GEMM_FAST) {
-      const bf16* src = x + m * cp.K + kg;
-      *reinterpret_cast<uint4*>(dst) = *reinterpret_cast<const uint4*>(src);
+      rg.c[c] = *reinterpret_cast<const uint4*>(x + m * cp.K + kg);
       continue;
     }
-    // implicit-GEMM gather: decode output position and patch element
     const int q = (int)(m % cp.Q);
     const int64_t np = m / cp.Q;
     const int p = (int)(np % cp.P);
@@ -101,14 +103,13 @@ DEV_INLINE void stage_a(const bf16* __restrict__ x, const ConvParams cp,
       const int h = p * cp.stride - cp.pad + r;
       const int w = q * cp.stride - cp.pad + s;
       if (h < 0 || h >= cp.H || w < 0 || w >= cp.W) {
-#pragma unroll
-        for (int j = 0; j < 8; ++j) dst[j] = 0;
+        rg.c[c] = make_uint4(0, 0, 0, 0);
       } else {
-        const bf16* src = x + (((int64_t)n * cp.H + h) * cp.W + w) * cp.Cin + ci;
-        *reinterpret_cast<uint4*>(dst) = *reinterpret_cast<const uint4*>(src);
+        rg.c[c] = *reinterpret_cast<const uint4*>(
+            x + (((int64_t)n * cp.H + h) * cp.W + w) * cp.Cin + ci);
       }
     } else {
-      // slow path (stem conv Cin=3 etc.): per-element gather
+      unsigned short tmp[8];
 #pragma unroll
       for (int j = 0; j < 8; ++j) {
         const int k = kg + j;
@@ -124,48 +125,72 @@ DEV_INLINE void stage_a(const bf16* __restrict__ x, const ConvParams cp,
             v = *reinterpret_cast<const unsigned short*>(
                 x + (((int64_t)n * cp.H + h) * cp.W + w) * cp.Cin + ci);
         }
-        dst[j] = v;
+        tmp[j] = v;
       }
+      rg.c[c] = *reinterpret_cast<const uint4*>(tmp);
     }
   }
 }
 
-DEV_INLINE void stage_b(const bf16* __restrict__ wgt, int ncols, int K,
-                        int n0, int k0, unsigned short* lds) {
+DEV_INLINE void load_b(const bf16* __restrict__ wgt, int ncols, int K,
+                       int n0, int k0, StageRegs& rg) {
   const int t = threadIdx.x;
 #pragma unroll
   for (int c = 0; c < 4; ++c) {
     const int idx = t + c * THREADS;
-    const int row = idx / (BK / 8);           // output-channel offset in tile
+    const int row = idx / (BK / 8);
     const int kc = (idx % (BK / 8)) * 8;
-    unsigned short* dst = lds + row * LDS_PITCH + kc;
     const int n = n0 + row;
     const int kg = k0 + kc;
     if (n >= ncols || kg >= K) {
-#pragma unroll
-      for (int j = 0; j < 8; ++j) dst[j] = 0;
+      rg.c[c] = make_uint4(0, 0, 0, 0);
     } else if (kg + 8 <= K) {
-      const bf16* src = wgt + (int64_t)n * K + kg;
-      *reinterpret_cast<uint4*>(dst) = *reinterpret_cast<const uint4*>(src);
+      rg.c[c] = *reinterpret_cast<const uint4*>(wgt + (int64_t)n * K + kg);
     } else {
+      unsigned short tmp[8];
 #pragma unroll
       for (int j = 0; j < 8; ++j)
-        dst[j] = (kg + j < K)
+        tmp[j] = (kg + j < K)
             ? *reinterpret_cast<const unsigned short*>(wgt + (int64_t)n * K + kg + j)
             : (unsigned short)0;
+      rg.c[c] = *reinterpret_cast<const uint4*>(tmp);
     }
+  }
+}
+
+DEV_INLINE void write_tile(const StageRegs& rg, unsigned short* lds) {
+  const int t = threadIdx.x;
+#pragma unroll
+  for (int c = 0; c < 4; ++c) {
+    const int idx = t + c * THREADS;
+    const int row = idx / (BK / 8);
+    const int kc = (idx % (BK / 8)) * 8;
+    *reinterpret_cast<uint4*>(lds + row * LDS_PITCH + kc) = rg.c[c];
   }
 }
 
 template <bool GEMM_FAST, bool RELU, bool HAS_BIAS>
 __global__ __launch_bounds__(THREADS, 2) void conv_implicit_gemm_kernel(
     const bf16* __restrict__ x, const bf16* __restrict__ wgt,
-    const float* __restrict__ bias, bf16* __restrict__ out, ConvParams cp) {
-  __shared__ unsigned short lds_a[BM * LDS_PITCH];
-  __shared__ unsigned short lds_b[BN * LDS_PITCH];
+    const float* __restrict__ bias, bf16* __restrict__ out, ConvParams cp,
+    int mtiles, int ntiles) {
+  __shared__ unsigned short lds_a[2][BM * LDS_PITCH];
+  __shared__ unsigned short lds_b[2][BN * LDS_PITCH];
 
-  const int64_t m0 = (int64_t)blockIdx.x * BM;
-  const int n0 = blockIdx.y * BN;
+  // XCD-aware block swizzle (guide T1, bijective variant): consecutive
+  // logical tiles land on the same XCD's L2 so neighbor tiles sharing A/B
+  // panels hit cache.
+  const int nwg = mtiles * ntiles;
+  int bid = blockIdx.x;
+  {
+    const int nx = 8;
+    const int qq = nwg / nx, rr = nwg % nx;
+    const int xcd = bid % nx, idx = bid / nx;
+    bid = (xcd < rr ? xcd * (qq + 1) : rr * (qq + 1) + (xcd - rr) * qq) + idx;
+  }
+  const int64_t m0 = (int64_t)(bid / ntiles) * BM;
+  const int n0 = (bid % ntiles) * BN;
+
   const int wave = threadIdx.x / 64;
   const int lane = threadIdx.x % 64;
   const int wm = (wave / 2) * 64;   // wave's 64x64 sub-tile origin
@@ -179,21 +204,33 @@ __global__ __launch_bounds__(THREADS, 2) void conv_implicit_gemm_kernel(
 
   const int frag_row = lane % 16;
   const int frag_koff = (lane / 16) * 8;
+  const int nk = (cp.K + BK - 1) / BK;
 
-  for (int k0 = 0; k0 < cp.K; k0 += BK) {
-    stage_a<GEMM_FAST>(x, cp, m0, k0, lds_a);
-    stage_b(wgt, cp.Cout, cp.K, n0, k0, lds_b);
-    __syncthreads();
+  StageRegs ra, rb;
+  load_a<GEMM_FAST>(x, cp, m0, 0, ra);
+  load_b(wgt, cp.Cout, cp.K, n0, 0, rb);
+  write_tile(ra, lds_a[0]);
+  write_tile(rb, lds_b[0]);
+  __syncthreads();
+
+  for (int t = 0; t < nk; ++t) {
+    const int cur = t & 1;
+    // issue next tile's global loads now — they stay in flight under the
+    // MFMA phase and are only waited for at the ds_write below
+    if (t + 1 < nk) {
+      load_a<GEMM_FAST>(x, cp, m0, (t + 1) * BK, ra);
+      load_b(wgt, cp.Cout, cp.K, n0, (t + 1) * BK, rb);
+    }
 #pragma unroll
     for (int ks = 0; ks < BK; ks += 32) {
       short8 afrag[4], bfrag[4];
 #pragma unroll
       for (int i = 0; i < 4; ++i) {
         const unsigned short* pa =
-            lds_a + (wm + i * 16 + frag_row) * LDS_PITCH + ks + frag_koff;
+            lds_a[cur] + (wm + i * 16 + frag_row) * LDS_PITCH + ks + frag_koff;
         afrag[i] = *reinterpret_cast<const short8*>(pa);
         const unsigned short* pb =
-            lds_b + (wn + i * 16 + frag_row) * LDS_PITCH + ks + frag_koff;
+            lds_b[cur] + (wn + i * 16 + frag_row) * LDS_PITCH + ks + frag_koff;
         bfrag[i] = *reinterpret_cast<const short8*>(pb);
       }
 #pragma unroll
@@ -203,7 +240,12 @@ __global__ __launch_bounds__(THREADS, 2) void conv_implicit_gemm_kernel(
           acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
               afrag[i], bfrag[j], acc[i][j], 0, 0, 0);
     }
-    __syncthreads();
+    if (t + 1 < nk) {
+      __syncthreads();  // everyone finished reading buf[cur^1] (tile t-1)
+      write_tile(ra, lds_a[cur ^ 1]);
+      write_tile(rb, lds_b[cur ^ 1]);
+      __syncthreads();
+    }
   }
 
   // epilogue: lane l, reg r -> row (l/16)*4 + r, col l%16 of each 16x16 frag
@@ -249,15 +291,16 @@ void mfma_gemm(Tensor a, Tensor bt, Tensor bias, Tensor out, bool relu,
   TORCH_CHECK(bt.size(1) == K && out.size(0) == M && out.size(1) == N);
   dwtmm::ConvParams cp{};
   cp.M = M; cp.K = K; cp.Cout = N;
-  dim3 grid((M + dwtmm::BM - 1) / dwtmm::BM, (N + dwtmm::BN - 1) / dwtmm::BN);
+  const int mtiles = (M + dwtmm::BM - 1) / dwtmm::BM;
+  const int ntiles = (N + dwtmm::BN - 1) / dwtmm::BN;
   auto run = [&](auto reluc, auto biasc) {
     hipLaunchKernelGGL(
         (dwtmm::conv_implicit_gemm_kernel<true, decltype(reluc)::value,
                                           decltype(biasc)::value>),
-        grid, dim3(dwtmm::THREADS), 0, dwtmm_stream(),
+        dim3(mtiles * ntiles), dim3(dwtmm::THREADS), 0, dwtmm_stream(),
         (const c10::BFloat16*)a.data_ptr(), (const c10::BFloat16*)bt.data_ptr(),
         has_bias ? bias.data_ptr<float>() : nullptr,
-        (c10::BFloat16*)out.data_ptr(), cp);
+        (c10::BFloat16*)out.data_ptr(), cp, mtiles, ntiles);
   };
   if (relu) { if (has_bias) run(std::true_type{}, std::true_type{});
               else run(std::true_type{}, std::false_type{}); }
@@ -278,8 +321,8 @@ void mfma_conv2d_fwd(Tensor x, Tensor wgt, Tensor bias, Tensor out,
   cp.KH = KH; cp.KW = KW; cp.stride = stride; cp.pad = pad;
   cp.M = N * P * Q;
   cp.K = KH * KW * Cin;
-  dim3 grid((cp.M + dwtmm::BM - 1) / dwtmm::BM,
-            (Cout + dwtmm::BN - 1) / dwtmm::BN);
+  const int mtiles = (cp.M + dwtmm::BM - 1) / dwtmm::BM;
+  const int ntiles = (Cout + dwtmm::BN - 1) / dwtmm::BN;
   const bool gemm_fast = (KH == 1 && KW == 1 && stride == 1 && pad == 0 &&
                           Cin % 8 == 0);
   auto run = [&](auto fastc, auto reluc, auto biasc) {
@@ -287,10 +330,10 @@ void mfma_conv2d_fwd(Tensor x, Tensor wgt, Tensor bias, Tensor out,
         (dwtmm::conv_implicit_gemm_kernel<decltype(fastc)::value,
                                           decltype(reluc)::value,
                                           decltype(biasc)::value>),
-        grid, dim3(dwtmm::THREADS), 0, dwtmm_stream(),
+        dim3(mtiles * ntiles), dim3(dwtmm::THREADS), 0, dwtmm_stream(),
         (const c10::BFloat16*)x.data_ptr(), (const c10::BFloat16*)wgt.data_ptr(),
         has_bias ? bias.data_ptr<float>() : nullptr,
-        (c10::BFloat16*)out.data_ptr(), cp);
+        (c10::BFloat16*)out.data_ptr(), cp, mtiles, ntiles);
   };
   auto pick_rb = [&](auto fastc) {
     if (relu) { if (has_bias) run(fastc, std::true_type{}, std::true_type{});
