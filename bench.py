@@ -100,7 +100,7 @@ def main():
         optimizer.zero_grad(set_to_none=True)
         out = model(data)
         s, t, td = torch.split(out, b, dim=0)
-        cls_loss = F.nll_loss(F.log_softmax(s.float(), dim=1), labels)
+        cls_loss = Fdwt.ce_loss(s, labels)
         mec = 0.1 * Fdwt.mec_loss(t, td)
         (cls_loss + mec).backward()
         if ddp.enabled:

@@ -38,7 +38,7 @@ def train_digits_epoch(args, model, device, source_loader, target_loader,
         chunks = torch.split(output, output.shape[0] // parts, dim=0)
         source_out = chunks[0]
 
-        cls_loss = F.nll_loss(F.log_softmax(source_out.float(), dim=1), source_y)
+        cls_loss = Fdwt.ce_loss(source_out, source_y)
         if loss_kind == "mec":
             aux = lambda_loss * Fdwt.mec_loss(chunks[1], chunks[2])
         else:

@@ -68,7 +68,7 @@ def train_infinite_collect_stats(args, model, device, source_train_loader,
         source_out, target_out, target_out_dup = torch.split(
             output, output.shape[0] // 3, dim=0)
 
-        cls_loss = F.nll_loss(F.log_softmax(source_out.float(), dim=1), source_y)
+        cls_loss = Fdwt.ce_loss(source_out, source_y)
         mec_loss = lambda_mec_loss * Fdwt.mec_loss(target_out, target_out_dup)
         loss = cls_loss + mec_loss
         loss.backward()

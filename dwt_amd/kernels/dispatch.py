@@ -87,3 +87,8 @@ def mec_loss(x, y):
 def entropy_loss(x):
     from . import hip_ops
     return hip_ops.entropy_loss(x)
+
+
+def ce_loss(x, target):
+    from . import hip_ops
+    return hip_ops.ce_loss(x, target)

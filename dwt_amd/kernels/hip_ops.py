@@ -396,3 +396,31 @@ def mec_loss(x, y):
 
 def entropy_loss(x):
     return _HipEntropyLoss.apply(x)
+
+
+class _HipCELoss(torch.autograd.Function):
+    """mean_n -log softmax(x)[n, target_n] (reference source loss)."""
+
+    @staticmethod
+    def forward(ctx, x, target):
+        ext = _ext()
+        x32 = x.detach().to(torch.float32).contiguous()
+        q = torch.empty_like(x32)
+        loss = torch.zeros(1, device=x.device, dtype=torch.float32)
+        ext.ce_fwd(x32, target.contiguous(), q, loss)
+        ctx.save_for_backward(q, target)
+        ctx.dtype = x.dtype
+        return loss[0]
+
+    @staticmethod
+    def backward(ctx, dloss):
+        ext = _ext()
+        q, target = ctx.saved_tensors
+        gscale = dloss.detach().to(torch.float32).reshape(1).contiguous()
+        dx = torch.empty_like(q)
+        ext.ce_bwd(q, target.contiguous(), gscale, dx)
+        return dx.to(ctx.dtype), None
+
+
+def ce_loss(x, target):
+    return _HipCELoss.apply(x, target)

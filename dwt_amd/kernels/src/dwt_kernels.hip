@@ -1610,6 +1610,39 @@ __global__ void entropy_fwd_kernel(
   }
 }
 
+// source cross-entropy: nll_loss(log_softmax(x)) as both reference loops
+// use it (usps_mnist.py:298, resnet50_dwt_mec_officehome.py:425)
+__global__ void ce_fwd_kernel(
+    const float* __restrict__ x, const int64_t* __restrict__ target,
+    float* __restrict__ q, float* __restrict__ loss, int N, int K) {
+  const int row = blockIdx.x * (blockDim.x / 64) + (threadIdx.x / 64);
+  const int lane = threadIdx.x & 63;
+  if (row >= N) return;
+  const float* xr = x + (int64_t)row * K;
+  float mx = -INFINITY;
+  for (int k = lane; k < K; k += 64) mx = fmaxf(mx, xr[k]);
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1) mx = fmaxf(mx, __shfl_down(mx, off, 64));
+  mx = __shfl(mx, 0, 64);
+  float ssum = 0.f;
+  for (int k = lane; k < K; k += 64) ssum += __expf(xr[k] - mx);
+  ssum = wave_reduce_sum(ssum);
+  ssum = __shfl(ssum, 0, 64);
+  const float lse = mx + __logf(ssum);
+  for (int k = lane; k < K; k += 64) q[(int64_t)row * K + k] = xr[k] - lse;
+  if (lane == 0) atomicAdd(loss, (lse - xr[target[row]]) / N);
+}
+
+__global__ void ce_bwd_kernel(
+    const float* __restrict__ q, const int64_t* __restrict__ target,
+    const float* __restrict__ gscale, float* __restrict__ dx, int N, int K) {
+  const int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= (int64_t)N * K) return;
+  const int row = i / K, k = i % K;
+  const float oh = (k == (int)target[row]) ? 1.f : 0.f;
+  dx[i] = (__expf(q[i]) - oh) * gscale[0] / N;
+}
+
 __global__ void entropy_bwd_kernel(
     const float* __restrict__ q, const float* __restrict__ hper,
     const float* __restrict__ gscale, float* __restrict__ dx, int N, int K) {
@@ -2271,6 +2304,25 @@ void entropy_fwd(Tensor x, Tensor q, Tensor hper, Tensor loss) {
                      hper.data_ptr<float>(), loss.data_ptr<float>(), N, K);
 }
 
+void ce_fwd(Tensor x, Tensor target, Tensor q, Tensor loss) {
+  const int N = x.size(0), K = x.size(1);
+  const int rows_per_block = 4;
+  hipLaunchKernelGGL(dwt::ce_fwd_kernel,
+                     dim3((N + rows_per_block - 1) / rows_per_block),
+                     dim3(rows_per_block * 64), 0, cur_stream(),
+                     x.data_ptr<float>(), target.data_ptr<int64_t>(),
+                     q.data_ptr<float>(), loss.data_ptr<float>(), N, K);
+}
+
+void ce_bwd(Tensor q, Tensor target, Tensor gscale, Tensor dx) {
+  const int N = q.size(0), K = q.size(1);
+  hipLaunchKernelGGL(dwt::ce_bwd_kernel,
+                     dim3(elementwise_blocks((int64_t)N * K, 256)), dim3(256),
+                     0, cur_stream(), q.data_ptr<float>(),
+                     target.data_ptr<int64_t>(), gscale.data_ptr<float>(),
+                     dx.data_ptr<float>(), N, K);
+}
+
 void entropy_bwd(Tensor q, Tensor hper, Tensor gscale, Tensor dx) {
   const int N = q.size(0), K = q.size(1);
   const int64_t total = (int64_t)N * K;
@@ -2322,5 +2374,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("mec_fwd", &mec_fwd);
   m.def("mec_bwd", &mec_bwd);
   m.def("entropy_fwd", &entropy_fwd);
+  m.def("ce_fwd", &ce_fwd);
+  m.def("ce_bwd", &ce_bwd);
   m.def("entropy_bwd", &entropy_bwd);
 }

@@ -557,3 +557,12 @@ class AddReluFn(torch.autograd.Function):
 
 def add_relu(a: torch.Tensor, b: torch.Tensor) -> torch.Tensor:
     return AddReluFn.apply(a, b)
+
+
+def ce_loss(logits: torch.Tensor, target: torch.Tensor) -> torch.Tensor:
+    """Source cross-entropy: nll_loss(log_softmax(x)) (usps_mnist.py:298)."""
+    if logits.is_cuda:
+        from ..kernels import dispatch
+        if dispatch.available():
+            return dispatch.ce_loss(logits, target)
+    return F.nll_loss(F.log_softmax(logits.float(), dim=1), target)

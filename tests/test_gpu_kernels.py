@@ -341,3 +341,23 @@ def test_add_relu_parity(dev):
         ref.backward(g)
         assert torch.allclose(a.grad.float(), a2.grad.float(), atol=1e-2)
         assert torch.allclose(b.grad.float(), b2.grad.float(), atol=1e-2)
+
+
+def test_ce_loss_parity(dev):
+    import torch.nn.functional as F
+    from dwt_amd.kernels import hip_ops
+    torch.manual_seed(14)
+    x = torch.randn(18, 65, device=dev, requires_grad=True)
+    t = torch.randint(0, 65, (18,), device=dev)
+    l_h = hip_ops.ce_loss(x, t)
+    ref = F.nll_loss(F.log_softmax(x.detach().float(), dim=1), t)
+    assert torch.allclose(l_h, ref, atol=1e-5)
+    l_h.backward()
+    x2 = x.detach().clone().requires_grad_(True)
+    F.nll_loss(F.log_softmax(x2, dim=1), t).backward()
+    assert torch.allclose(x.grad, x2.grad, atol=1e-6)
+    # bf16 logits path
+    xb = torch.randn(6, 10, device=dev, dtype=torch.bfloat16, requires_grad=True)
+    tb = torch.randint(0, 10, (6,), device=dev)
+    hip_ops.ce_loss(xb, tb).backward()
+    assert xb.grad is not None
