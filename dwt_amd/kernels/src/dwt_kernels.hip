@@ -1405,6 +1405,97 @@ __global__ void ema_update_kernel(float* __restrict__ rm,
 }
 
 // ===========================================================================
+// Pooling, NHWC.
+// maxpool KxK/stride/pad fwd saves the argmax window index; backward gathers
+// (each input cell checks the <=ceil(K/s)^2 windows that contain it) — no
+// atomics.  Global average pool reduces all positions per (n, c).
+// ===========================================================================
+
+template <typename T>
+__global__ void maxpool_nhwc_fwd_kernel(
+    const T* __restrict__ x, T* __restrict__ out, uint8_t* __restrict__ idx,
+    int C, int H, int W, int P, int Q, int KS, int stride, int pad,
+    int64_t total) {  // total = N*P*Q*C
+  const int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= total) return;
+  const int c = i % C;
+  int64_t r = i / C;
+  const int q = r % Q; r /= Q;
+  const int p = r % P; r /= P;
+  const int n = (int)r;
+  float best = -INFINITY;
+  int bidx = 0;
+  for (int kh = 0; kh < KS; ++kh) {
+    const int h = p * stride - pad + kh;
+    if (h < 0 || h >= H) continue;
+    for (int kw = 0; kw < KS; ++kw) {
+      const int w = q * stride - pad + kw;
+      if (w < 0 || w >= W) continue;
+      const float v = ldf(x + (((int64_t)n * H + h) * W + w) * C + c);
+      if (v > best) { best = v; bidx = kh * KS + kw; }
+    }
+  }
+  stf(out + i, best);
+  idx[i] = (uint8_t)bidx;
+}
+
+template <typename T>
+__global__ void maxpool_nhwc_bwd_kernel(
+    const T* __restrict__ dout, const uint8_t* __restrict__ idx,
+    T* __restrict__ dx, int C, int H, int W, int P, int Q, int KS, int stride,
+    int pad, int64_t total_in) {  // total_in = N*H*W*C
+  const int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= total_in) return;
+  const int c = i % C;
+  int64_t r = i / C;
+  const int w = r % W; r /= W;
+  const int h = r % H; r /= H;
+  const int n = (int)r;
+  float acc = 0.f;
+  // windows (p, q) with p*stride - pad <= h < p*stride - pad + KS
+  const int pmin = max(0, (h + pad - KS + stride) / stride);
+  const int pmax = min(P - 1, (h + pad) / stride);
+  const int qmin = max(0, (w + pad - KS + stride) / stride);
+  const int qmax = min(Q - 1, (w + pad) / stride);
+  for (int p = pmin; p <= pmax; ++p) {
+    const int kh = h - (p * stride - pad);
+    if (kh < 0 || kh >= KS) continue;
+    for (int q = qmin; q <= qmax; ++q) {
+      const int kw = w - (q * stride - pad);
+      if (kw < 0 || kw >= KS) continue;
+      const int64_t o = (((int64_t)n * P + p) * Q + q) * C + c;
+      if (idx[o] == (uint8_t)(kh * KS + kw)) acc += ldf(dout + o);
+    }
+  }
+  stf(dx + i, acc);
+}
+
+template <typename T>
+__global__ void gap_nhwc_fwd_kernel(const T* __restrict__ x,
+                                    T* __restrict__ out, int C, int64_t HW,
+                                    int N) {
+  // one wave per (n, c-chunk of 64): lanes cover 64 channels, loop positions
+  const int n = blockIdx.y;
+  const int c = blockIdx.x * 64 + (threadIdx.x & 63);
+  if (c >= C) return;
+  const T* base = x + (int64_t)n * HW * C + c;
+  float s = 0.f;
+  for (int64_t m = 0; m < HW; ++m) s += ldf(base + m * C);
+  stf(out + (int64_t)n * C + c, s / (float)HW);
+}
+
+template <typename T>
+__global__ void gap_nhwc_bwd_kernel(const T* __restrict__ dout,
+                                    T* __restrict__ dx, int C, int64_t HW,
+                                    int64_t total) {
+  const int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= total) return;
+  const int c = i % C;
+  const int64_t n = i / (HW * C);
+  stf(dx + i, ldf(dout + n * C + c) / (float)HW);
+}
+
+// ===========================================================================
 // Fused residual join: out = relu(a + b); backward: da = db = dout*(out>0)
 // (the reference's `relu(out + identity)` at every bottleneck exit,
 // resnet50_dwt_mec_officehome.py:239-240)
@@ -2183,6 +2274,53 @@ void bn_bwd_apply(Tensor x, Tensor dout, Tensor out, Tensor mean, Tensor istd,
   });
 }
 
+void maxpool_cl_fwd(Tensor x, Tensor out, Tensor idx, int64_t C, int64_t H,
+                    int64_t W, int64_t P, int64_t Q, int64_t KS,
+                    int64_t stride, int64_t pad) {
+  const int64_t total = out.numel();
+  DISPATCH_FT(x, "maxpool_cl_fwd", [&] {
+    hipLaunchKernelGGL((dwt::maxpool_nhwc_fwd_kernel<scalar_t>),
+                       dim3(elementwise_blocks(total, 256)), dim3(256), 0,
+                       cur_stream(), x.data_ptr<scalar_t>(),
+                       out.data_ptr<scalar_t>(), idx.data_ptr<uint8_t>(),
+                       (int)C, (int)H, (int)W, (int)P, (int)Q, (int)KS,
+                       (int)stride, (int)pad, total);
+  });
+}
+
+void maxpool_cl_bwd(Tensor dout, Tensor idx, Tensor dx, int64_t C, int64_t H,
+                    int64_t W, int64_t P, int64_t Q, int64_t KS,
+                    int64_t stride, int64_t pad) {
+  const int64_t total_in = dx.numel();
+  DISPATCH_FT(dout, "maxpool_cl_bwd", [&] {
+    hipLaunchKernelGGL((dwt::maxpool_nhwc_bwd_kernel<scalar_t>),
+                       dim3(elementwise_blocks(total_in, 256)), dim3(256), 0,
+                       cur_stream(), dout.data_ptr<scalar_t>(),
+                       idx.data_ptr<uint8_t>(), dx.data_ptr<scalar_t>(),
+                       (int)C, (int)H, (int)W, (int)P, (int)Q, (int)KS,
+                       (int)stride, (int)pad, total_in);
+  });
+}
+
+void gap_cl_fwd(Tensor x, Tensor out, int64_t C, int64_t HW, int64_t N) {
+  DISPATCH_FT(x, "gap_cl_fwd", [&] {
+    dim3 grid((C + 63) / 64, N);
+    hipLaunchKernelGGL((dwt::gap_nhwc_fwd_kernel<scalar_t>), grid, dim3(64), 0,
+                       cur_stream(), x.data_ptr<scalar_t>(),
+                       out.data_ptr<scalar_t>(), (int)C, HW, (int)N);
+  });
+}
+
+void gap_cl_bwd(Tensor dout, Tensor dx, int64_t C, int64_t HW) {
+  const int64_t total = dx.numel();
+  DISPATCH_FT(dout, "gap_cl_bwd", [&] {
+    hipLaunchKernelGGL((dwt::gap_nhwc_bwd_kernel<scalar_t>),
+                       dim3(elementwise_blocks(total, 256)), dim3(256), 0,
+                       cur_stream(), dout.data_ptr<scalar_t>(),
+                       dx.data_ptr<scalar_t>(), (int)C, HW, total);
+  });
+}
+
 void add_relu_fwd(Tensor a, Tensor b, Tensor out) {
   const int64_t n = a.numel();
   DISPATCH_FT(a, "add_relu_fwd", [&] {
@@ -2367,6 +2505,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("bn_bwd_reduce", &bn_bwd_reduce);
   m.def("bn_bwd_apply", &bn_bwd_apply);
   m.def("ema_update", &ema_update);
+  m.def("maxpool_cl_fwd", &maxpool_cl_fwd);
+  m.def("maxpool_cl_bwd", &maxpool_cl_bwd);
+  m.def("gap_cl_fwd", &gap_cl_fwd);
+  m.def("gap_cl_bwd", &gap_cl_bwd);
   m.def("add_relu_fwd", &add_relu_fwd);
   m.def("add_relu_bwd", &add_relu_bwd);
   m.def("fused_sgd", &fused_sgd);

@@ -20,6 +20,7 @@ import torch.nn.functional as F
 
 from ..ops.batch_norm import DomainBatchNorm1d
 from ..ops.whitening import WTransform2d
+from ..ops.pooling import max_pool2d
 from .sites import norm_site
 
 
@@ -83,7 +84,7 @@ class LeNet(nn.Module):
         for conv, branches, gamma, beta in w_sites:
             x = conv(x)
             x = norm_site(x, branches, gamma, beta, training=tr, relu=True)
-            x = F.max_pool2d(x, kernel_size=2, stride=2)
+            x = max_pool2d(x, kernel_size=2, stride=2)
 
         x = x.reshape(x.shape[0], -1)
         fc_sites = [

@@ -26,6 +26,7 @@ import torch.nn as nn
 from ..ops.batch_norm import DomainBatchNorm2d
 from ..ops.whitening import WhiteningScaleShift
 from ..ops.functional import add_relu
+from ..ops.pooling import MaxPool2dDWT, global_avg_pool
 from .sites import norm_site
 
 
@@ -148,7 +149,7 @@ class ResNetDWT(nn.Module):
         self.conv1 = nn.Conv2d(3, 64, kernel_size=7, stride=2, padding=3, bias=False)
         (self.bns1, self.bnt1, self.bnt1_aug), self.gamma1, self.beta1 = \
             _make_wh_branches(64, group_size, bn_dict, "bn1", whiten_mode)
-        self.maxpool = nn.MaxPool2d(kernel_size=3, stride=2, padding=1)
+        self.maxpool = MaxPool2dDWT(kernel_size=3, stride=2, padding=1)
         self.layer1 = self._make_layer(block, 64, layers[0], bn_dict, layer=1,
                                        group_size=group_size)
         self.layer2 = self._make_layer(block, 128, layers[1], bn_dict, layer=2,
@@ -185,8 +186,7 @@ class ResNetDWT(nn.Module):
         x = self.layer2(x)
         x = self.layer3(x)
         x = self.layer4(x)
-        x = self.avgpool(x)
-        x = x.reshape(x.size(0), -1)
+        x = global_avg_pool(x)
         return self.fc_out(x)
 
 

@@ -361,3 +361,32 @@ def test_ce_loss_parity(dev):
     tb = torch.randint(0, 10, (6,), device=dev)
     hip_ops.ce_loss(xb, tb).backward()
     assert xb.grad is not None
+
+
+def test_pooling_parity(dev):
+    import torch.nn.functional as F
+    from dwt_amd.ops.pooling import MaxPool2dFn, GlobalAvgPoolFn
+    torch.manual_seed(15)
+    for ks, st, pad, shape in [(3, 2, 1, (4, 64, 15, 15)),
+                               (2, 2, 0, (4, 32, 28, 28))]:
+        x = torch.randn(*shape, device=dev, dtype=torch.bfloat16) \
+            .contiguous(memory_format=torch.channels_last).requires_grad_(True)
+        x2 = x.detach().clone().requires_grad_(True)
+        out = MaxPool2dFn.apply(x, ks, st, pad)
+        ref = F.max_pool2d(x2, kernel_size=ks, stride=st, padding=pad)
+        assert torch.allclose(out.float(), ref.float(), atol=1e-2), (ks, st)
+        g = torch.randn_like(ref)
+        out.backward(g.contiguous(memory_format=torch.channels_last))
+        ref.backward(g)
+        assert torch.allclose(x.grad.float(), x2.grad.float(), atol=1e-2)
+
+    x = torch.randn(6, 256, 7, 7, device=dev, dtype=torch.bfloat16) \
+        .contiguous(memory_format=torch.channels_last).requires_grad_(True)
+    x2 = x.detach().clone().requires_grad_(True)
+    out = GlobalAvgPoolFn.apply(x)
+    ref = F.adaptive_avg_pool2d(x2, (1, 1)).reshape(6, 256)
+    assert torch.allclose(out.float(), ref.float(), atol=1e-2)
+    g = torch.randn_like(ref)
+    out.backward(g)
+    ref.backward(g)
+    assert torch.allclose(x.grad.float(), x2.grad.float(), atol=1e-2)
