@@ -3,6 +3,8 @@ gather backward — no atomics) and global average pool, with transparent
 fallback to torch pooling off-GPU / off-channels_last."""
 from __future__ import annotations
 
+import os
+
 import torch
 import torch.nn as nn
 import torch.nn.functional as F
@@ -14,6 +16,8 @@ def _cl4(t):
 
 
 def _hip_ok(x):
+    if os.environ.get("DWT_AMD_POOL") == "torch":
+        return False
     if not x.is_cuda or x.dtype not in (torch.float32, torch.bfloat16):
         return False
     from ..kernels import dispatch

@@ -145,6 +145,8 @@ def main(argv=None):
                            optimizer, epoch, args.lambda_entropy_loss,
                            loss_kind=args.loss, logger=logger,
                            grad_sync=ddp.sync if ddp.enabled else None)
+        if world > 1:
+            ddp.broadcast_buffers()
         if rank == 0:
             test(args, model, device, test_loader, logger=logger)
         if rank == 0 and args.checkpoint_path:
