@@ -29,11 +29,11 @@ def _whiten_layout(x, c):
 
 
 def _bn_layout(x, c):
-    if x.dim() == 2 and c % 8 == 0 and x.is_contiguous():
+    if x.dim() == 2 and c % 4 == 0 and x.is_contiguous():
         return "cl"  # (N, C) IS channels-last: C contiguous per position
     if x.dim() == 4 and x.is_contiguous(memory_format=torch.channels_last) \
             and not x.is_contiguous():
-        if c % 8 == 0 and (c <= 1024 or c % 1024 == 0):
+        if c % 4 == 0 and (c <= 1024 or c % 1024 == 0):
             return "cl"
     return "nchw"
 

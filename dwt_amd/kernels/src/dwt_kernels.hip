@@ -1055,7 +1055,7 @@ __global__ void whiten_bwd_apply_nhwc_kernel(
 template <typename T>
 __global__ void bn_stats_nhwc_kernel(
     const T* __restrict__ x, float* __restrict__ acc, int C, int64_t M) {
-  constexpr int VC = 8;
+  constexpr int VC = 4;
   const int CW = C < 1024 ? C : 1024;
   const int NCH = CW / VC;
   const int c0 = blockIdx.z * 1024 + (threadIdx.x % NCH) * VC;
@@ -1095,7 +1095,7 @@ __global__ void bn_apply_nhwc_kernel(
     const float* __restrict__ istd, const T* __restrict__ gamma,
     const T* __restrict__ beta, T* __restrict__ out, int C, int64_t M,
     int relu, int has_affine) {
-  constexpr int VC = 8;
+  constexpr int VC = 4;
   const int CW = C < 1024 ? C : 1024;
   const int NCH = CW / VC;
   const int c0 = blockIdx.z * 1024 + (threadIdx.x % NCH) * VC;
@@ -1129,7 +1129,7 @@ __global__ void bn_bwd_reduce_nhwc_kernel(
     const T* __restrict__ out, const float* __restrict__ mean,
     const float* __restrict__ istd, float* __restrict__ sums, int C,
     int64_t M, int relu) {
-  constexpr int VC = 8;
+  constexpr int VC = 4;
   const int CW = C < 1024 ? C : 1024;
   const int NCH = CW / VC;
   const int c0 = blockIdx.z * 1024 + (threadIdx.x % NCH) * VC;
@@ -1178,7 +1178,7 @@ __global__ void bn_bwd_apply_nhwc_kernel(
     const float* __restrict__ istd, const T* __restrict__ gamma,
     const float* __restrict__ sums, T* __restrict__ dx, int C, int64_t M,
     float inv_m, int relu, int has_affine, int use_batch) {
-  constexpr int VC = 8;
+  constexpr int VC = 4;
   const int CW = C < 1024 ? C : 1024;
   const int NCH = CW / VC;
   const int c0 = blockIdx.z * 1024 + (threadIdx.x % NCH) * VC;
@@ -1945,7 +1945,7 @@ void bn_stats_cl(Tensor x, Tensor acc, Tensor mean, Tensor istd, Tensor var_unb,
                  int64_t C, int64_t M, double eps) {
   const int zslices = (C + 1023) / 1024;
   DISPATCH_FT(x, "bn_stats_cl", [&] {
-    const int NCH = (C < 1024 ? C : 1024) / 8;
+    const int NCH = (C < 1024 ? C : 1024) / 4;
     dim3 grid(nhwc_reduce_blocks(M, 256 / NCH > 0 ? 256 / NCH : 1, zslices), 1,
               zslices);
     hipLaunchKernelGGL((dwt::bn_stats_nhwc_kernel<scalar_t>), grid, dim3(256),
@@ -1964,7 +1964,7 @@ void bn_apply_cl(Tensor x, Tensor mean, Tensor istd, Tensor gamma, Tensor beta,
                  Tensor out, int64_t C, int64_t M, bool relu, bool has_affine) {
   const int zslices = (C + 1023) / 1024;
   DISPATCH_FT(x, "bn_apply_cl", [&] {
-    const int NCH = (C < 1024 ? C : 1024) / 8;
+    const int NCH = (C < 1024 ? C : 1024) / 4;
     const int rpi = std::max(256 / NCH, 1);
     dim3 grid(nhwc_elem_blocks(M, rpi, zslices), 1, zslices);
     hipLaunchKernelGGL((dwt::bn_apply_nhwc_kernel<scalar_t>), grid,
@@ -1982,7 +1982,7 @@ void bn_bwd_reduce_cl(Tensor x, Tensor dout, Tensor out, Tensor mean,
                       bool relu) {
   const int zslices = (C + 1023) / 1024;
   DISPATCH_FT(x, "bn_bwd_reduce_cl", [&] {
-    const int NCH = (C < 1024 ? C : 1024) / 8;
+    const int NCH = (C < 1024 ? C : 1024) / 4;
     dim3 grid(nhwc_reduce_blocks(M, 256 / NCH > 0 ? 256 / NCH : 1, zslices), 1,
               zslices);
     hipLaunchKernelGGL((dwt::bn_bwd_reduce_nhwc_kernel<scalar_t>), grid,
@@ -1999,7 +1999,7 @@ void bn_bwd_apply_cl(Tensor x, Tensor dout, Tensor out, Tensor mean,
                      bool use_batch) {
   const int zslices = (C + 1023) / 1024;
   DISPATCH_FT(x, "bn_bwd_apply_cl", [&] {
-    const int NCH = (C < 1024 ? C : 1024) / 8;
+    const int NCH = (C < 1024 ? C : 1024) / 4;
     const int rpi = std::max(256 / NCH, 1);
     dim3 grid(nhwc_elem_blocks(M, rpi, zslices), 1, zslices);
     hipLaunchKernelGGL((dwt::bn_bwd_apply_nhwc_kernel<scalar_t>), grid,
