@@ -47,8 +47,9 @@ struct ConvParams {
   int N, H, W, Cin;      // input
   int P, Q, Cout;        // output spatial + channels
   int KH, KW, stride, pad;
-  int64_t M;             // N*P*Q
-  int K;                 // KH*KW*Cin
+  int64_t M;             // N*P*Q (fwd) / N*H*W (dgrad)
+  int K;                 // KH*KW*Cin (fwd) / KH*KW*Cdy (dgrad)
+  int Cdy;               // dgrad: channels of dy
 };
 
 DEV_INLINE float bf16_to_f(unsigned short u) {
@@ -72,7 +73,10 @@ struct StageRegs {
   uint4 c[4];
 };
 
-template <bool GEMM_FAST>  // true: A is a plain dense [M][K] matrix
+// A-operand gather modes
+enum AMode { A_DENSE = 0, A_CONV = 1, A_DGRAD = 2 };
+
+template <int MODE>
 DEV_INLINE void load_a(const bf16* __restrict__ x, const ConvParams& cp,
                        int64_t m0, int k0, StageRegs& rg) {
   const int t = threadIdx.x;
@@ -87,8 +91,45 @@ DEV_INLINE void load_a(const bf16* __restrict__ x, const ConvParams& cp,
       rg.c[c] = make_uint4(0, 0, 0, 0);
       continue;
     }
-    if (GEMM_FAST) {
+    if (MODE == A_DENSE) {
       rg.c[c] = *reinterpret_cast<const uint4*>(x + m * cp.K + kg);
+      continue;
+    }
+    if (MODE == A_DGRAD) {
+      // x here is dy (N,P,Q,Cdy) raw NHWC; m -> input position (n,h,w);
+      // k -> (r, s, co) with co fastest.  dx[n,h,w,ci] needs
+      // dy[n, (h+pad-r)/stride, (w+pad-s)/stride, co] when divisible.
+      const int wi = (int)(m % cp.W);
+      const int64_t nh = m / cp.W;
+      const int hi = (int)(nh % cp.H);
+      const int n = (int)(nh / cp.H);
+      const int co = kg % cp.Cdy;
+      const int rs = kg / cp.Cdy;
+      const int sx = rs % cp.KW;
+      const int r = rs / cp.KW;
+      const int hp = hi + cp.pad - r;
+      const int wp = wi + cp.pad - sx;
+      bool ok = hp >= 0 && wp >= 0 && hp % cp.stride == 0 && wp % cp.stride == 0;
+      const int pp = hp / cp.stride, qq = wp / cp.stride;
+      ok = ok && pp < cp.P && qq < cp.Q;
+      if (!ok) {
+        rg.c[c] = make_uint4(0, 0, 0, 0);
+      } else if (cp.Cdy % 8 == 0) {
+        rg.c[c] = *reinterpret_cast<const uint4*>(
+            x + (((int64_t)n * cp.P + pp) * cp.Q + qq) * cp.Cdy + co);
+      } else {
+        unsigned short tmp[8];
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          const int k = kg + j;
+          unsigned short v = 0;
+          if (k < cp.K && k / cp.Cdy == rs)
+            v = *reinterpret_cast<const unsigned short*>(
+                x + (((int64_t)n * cp.P + pp) * cp.Q + qq) * cp.Cdy + k % cp.Cdy);
+          tmp[j] = v;
+        }
+        rg.c[c] = *reinterpret_cast<const uint4*>(tmp);
+      }
       continue;
     }
     const int q = (int)(m % cp.Q);
@@ -174,7 +215,7 @@ DEV_INLINE void write_tile(const StageRegs& rg, unsigned short* lds) {
 // single-buffer loop wins for short-K and for the implicit gather path
 // (within-shape A/B on the R50 shapes).  SWZ: XCD-aware bijective block
 // remap (guide T1) — only when the grid has several N-tiles to share.
-template <bool GEMM_FAST, bool RELU, bool HAS_BIAS, bool PIPE, bool SWZ>
+template <int MODE, bool RELU, bool HAS_BIAS, bool PIPE, bool SWZ>
 __global__ __launch_bounds__(THREADS, 2) void conv_implicit_gemm_kernel(
     const bf16* __restrict__ x, const bf16* __restrict__ wgt,
     const float* __restrict__ bias, bf16* __restrict__ out, ConvParams cp,
@@ -210,7 +251,7 @@ __global__ __launch_bounds__(THREADS, 2) void conv_implicit_gemm_kernel(
 
   StageRegs ra, rb;
   if (PIPE) {
-    load_a<GEMM_FAST>(x, cp, m0, 0, ra);
+    load_a<MODE>(x, cp, m0, 0, ra);
     load_b(wgt, cp.Cout, cp.K, n0, 0, rb);
     write_tile(ra, lds_a[0]);
     write_tile(rb, lds_b[0]);
@@ -223,11 +264,11 @@ __global__ __launch_bounds__(THREADS, 2) void conv_implicit_gemm_kernel(
       // issue next tile's global loads now — they stay in flight under the
       // MFMA phase and are only waited for at the ds_write below
       if (t + 1 < nk) {
-        load_a<GEMM_FAST>(x, cp, m0, (t + 1) * BK, ra);
+        load_a<MODE>(x, cp, m0, (t + 1) * BK, ra);
         load_b(wgt, cp.Cout, cp.K, n0, (t + 1) * BK, rb);
       }
     } else {
-      load_a<GEMM_FAST>(x, cp, m0, t * BK, ra);
+      load_a<MODE>(x, cp, m0, t * BK, ra);
       load_b(wgt, cp.Cout, cp.K, n0, t * BK, rb);
       write_tile(ra, lds_a[0]);
       write_tile(rb, lds_b[0]);
@@ -313,7 +354,8 @@ void mfma_gemm(Tensor a, Tensor bt, Tensor bias, Tensor out, bool relu,
   const bool swz = ntiles >= 8 && mtiles * ntiles >= 16;
   auto run3 = [&](auto reluc, auto biasc, auto pipec, auto swzc) {
     hipLaunchKernelGGL(
-        (dwtmm::conv_implicit_gemm_kernel<true, decltype(reluc)::value,
+        (dwtmm::conv_implicit_gemm_kernel<dwtmm::A_DENSE,
+                                          decltype(reluc)::value,
                                           decltype(biasc)::value,
                                           decltype(pipec)::value,
                                           decltype(swzc)::value>),
@@ -358,7 +400,8 @@ void mfma_conv2d_fwd(Tensor x, Tensor wgt, Tensor bias, Tensor out,
   auto run = [&](auto fastc, auto reluc, auto biasc) {
     auto launch = [&](auto pipec, auto swzc) {
       hipLaunchKernelGGL(
-          (dwtmm::conv_implicit_gemm_kernel<decltype(fastc)::value,
+          (dwtmm::conv_implicit_gemm_kernel<decltype(fastc)::value ? dwtmm::A_DENSE
+                                                                   : dwtmm::A_CONV,
                                             decltype(reluc)::value,
                                             decltype(biasc)::value,
                                             decltype(pipec)::value,
@@ -381,4 +424,28 @@ void mfma_conv2d_fwd(Tensor x, Tensor wgt, Tensor bias, Tensor out,
   };
   if (gemm_fast) pick_rb(std::true_type{});
   else pick_rb(std::false_type{});
+}
+
+
+// dgrad: dx (N,H,W,Cin) = implicit-GEMM over dy (N,P,Q,Cdy) with
+// wd (Cin, KH*KW*Cdy) = weight permuted to [ci][r][s][co] (host side).
+void mfma_conv2d_dgrad(Tensor dy, Tensor wd, Tensor dx, int64_t N, int64_t H,
+                       int64_t W, int64_t Cin, int64_t P, int64_t Q,
+                       int64_t Cdy, int64_t KH, int64_t KW, int64_t stride,
+                       int64_t pad) {
+  dwtmm::ConvParams cp{};
+  cp.N = N; cp.H = H; cp.W = W; cp.Cin = Cin;
+  cp.P = P; cp.Q = Q; cp.Cout = Cin;  // GEMM output channels = Cin
+  cp.KH = KH; cp.KW = KW; cp.stride = stride; cp.pad = pad;
+  cp.Cdy = Cdy;
+  cp.M = N * H * W;
+  cp.K = KH * KW * Cdy;
+  const int mtiles = (cp.M + dwtmm::BM - 1) / dwtmm::BM;
+  const int ntiles = (Cin + dwtmm::BN - 1) / dwtmm::BN;
+  hipLaunchKernelGGL(
+      (dwtmm::conv_implicit_gemm_kernel<dwtmm::A_DGRAD, false, false, false,
+                                        false>),
+      dim3(mtiles * ntiles), dim3(dwtmm::THREADS), 0, dwtmm_stream(),
+      (const c10::BFloat16*)dy.data_ptr(), (const c10::BFloat16*)wd.data_ptr(),
+      nullptr, (c10::BFloat16*)dx.data_ptr(), cp, mtiles, ntiles);
 }

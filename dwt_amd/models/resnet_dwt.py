@@ -30,12 +30,20 @@ from ..ops.pooling import MaxPool2dDWT, global_avg_pool
 from .sites import norm_site
 
 
+def _conv_cls():
+    import os
+    if os.environ.get("DWT_AMD_CONV") == "hip":
+        from ..ops.mfma import MFMAConv2d
+        return MFMAConv2d
+    return nn.Conv2d
+
+
 def conv3x3(cin, cout, stride=1):
-    return nn.Conv2d(cin, cout, kernel_size=3, stride=stride, padding=1, bias=False)
+    return _conv_cls()(cin, cout, kernel_size=3, stride=stride, padding=1, bias=False)
 
 
 def conv1x1(cin, cout, stride=1):
-    return nn.Conv2d(cin, cout, kernel_size=1, stride=stride, bias=False)
+    return _conv_cls()(cin, cout, kernel_size=1, stride=stride, bias=False)
 
 
 def _get(bn_dict: Optional[Dict], key: str):

@@ -106,3 +106,62 @@ def conv2d_fwd(x: torch.Tensor, weight: torch.Tensor,
     _ext().mfma_conv2d_fwd(x, wgt, b32, out, n, h, w, cin, p, q, cout,
                            kh, kw, stride, padding, relu, has_bias)
     return out
+
+
+def conv2d_dgrad(dy: torch.Tensor, weight: torch.Tensor, in_shape,
+                 stride: int = 1, padding: int = 0) -> torch.Tensor:
+    """dx for an NHWC conv: implicit GEMM over dy with the weight permuted to
+    [ci][r][s][co] (tiny host-side permute per call)."""
+    n, cin, h, w = in_shape
+    cout, _, kh, kw = weight.shape
+    p, q = dy.shape[2], dy.shape[3]
+    dyc = dy.contiguous(memory_format=torch.channels_last)
+    # CL storage of weight is [co][kh][kw][ci]; build [ci][kh][kw][co]
+    wv = weight.permute(0, 2, 3, 1)              # (Cout, KH, KW, Cin), contig view of CL
+    wd = wv.permute(3, 1, 2, 0).contiguous()     # (Cin, KH, KW, Cout)
+    dx = torch.empty(n, cin, h, w, device=dy.device, dtype=torch.bfloat16,
+                     memory_format=torch.channels_last)
+    _ext().mfma_conv2d_dgrad(dyc, wd, dx, n, h, w, cin, p, q, cout,
+                             kh, kw, stride, padding)
+    return dx
+
+
+class _MFMAConv2dFn(torch.autograd.Function):
+    """Training conv: our MFMA implicit-GEMM fwd + dgrad; wgrad via the
+    library (MFMA wgrad kernel is the round-2 item)."""
+
+    @staticmethod
+    def forward(ctx, x, weight, bias, stride, padding):
+        out = conv2d_fwd(x, weight, bias=bias, stride=stride, padding=padding)
+        ctx.save_for_backward(x, weight)
+        ctx.meta = (stride, padding, bias is not None)
+        return out
+
+    @staticmethod
+    def backward(ctx, dout):
+        x, weight = ctx.saved_tensors
+        stride, padding, has_bias = ctx.meta
+        dout = dout.contiguous(memory_format=torch.channels_last)
+        dx = conv2d_dgrad(dout, weight, x.shape, stride=stride, padding=padding)
+        _, dw, db = torch.ops.aten.convolution_backward(
+            dout, x, weight, [weight.shape[0]] if has_bias else None,
+            [stride, stride], [padding, padding], [1, 1], False, [0, 0], 1,
+            [False, True, has_bias])
+        return dx, dw, db, None, None
+
+
+class MFMAConv2d(nn.Conv2d):
+    """Conv2d on the hand-written MFMA kernels (fwd + dgrad) when running
+    bf16 channels_last on GPU; F.conv2d otherwise.  Enable in the models with
+    DWT_AMD_CONV=hip."""
+
+    def forward(self, x):
+        if x.is_cuda and x.dtype == torch.bfloat16 and x.shape[1] >= 8 \
+                and x.is_contiguous(memory_format=torch.channels_last) \
+                and self.stride[0] == self.stride[1] \
+                and self.padding[0] == self.padding[1]:
+            from ..kernels import dispatch
+            if dispatch.available():
+                return _MFMAConv2dFn.apply(x, self.weight, self.bias,
+                                           self.stride[0], self.padding[0])
+        return super().forward(x)

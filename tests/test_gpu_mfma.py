@@ -97,3 +97,45 @@ def test_mfma_conv_fwd_vs_miopen(dev, shape):
     err = _rel_err(ours, ref)
     assert err < 0.05, (shape, err)
     assert ours.is_contiguous(memory_format=torch.channels_last)
+
+
+@pytest.mark.parametrize("shape", [
+    (4, 64, 28, 28, 64, 1, 1, 0),
+    (4, 64, 28, 28, 64, 3, 1, 1),
+    (4, 128, 28, 28, 128, 3, 2, 1),
+])
+def test_mfma_conv_dgrad_vs_miopen(dev, shape):
+    torch.manual_seed(3)
+    from dwt_amd.ops.mfma import conv2d_dgrad
+    n, cin, h, w, cout, k, stride, pad = shape
+    x = torch.randn(n, cin, h, w, device=dev).to(torch.bfloat16) \
+        .contiguous(memory_format=torch.channels_last).requires_grad_(True)
+    wt = (torch.randn(cout, cin, k, k, device=dev) / (k * cin ** 0.5)) \
+        .to(torch.bfloat16).contiguous(memory_format=torch.channels_last)
+    out = F.conv2d(x, wt, stride=stride, padding=pad)
+    g = torch.randn_like(out).contiguous(memory_format=torch.channels_last)
+    out.backward(g)
+    dx_ref = x.grad
+    dx = conv2d_dgrad(g, wt, x.shape, stride=stride, padding=pad)
+    assert _rel_err(dx, dx_ref) < 0.05, (shape, _rel_err(dx, dx_ref))
+
+
+def test_mfma_conv2d_module_trains(dev):
+    from dwt_amd.ops.mfma import MFMAConv2d
+    torch.manual_seed(4)
+    conv = MFMAConv2d(64, 128, kernel_size=3, stride=1, padding=1, bias=False)
+    conv = conv.to(dev).to(torch.bfloat16).to(memory_format=torch.channels_last)
+    ref = torch.nn.Conv2d(64, 128, 3, 1, 1, bias=False)
+    ref = ref.to(dev).to(torch.bfloat16).to(memory_format=torch.channels_last)
+    with torch.no_grad():
+        ref.weight.copy_(conv.weight)
+    x1 = torch.randn(4, 64, 16, 16, device=dev, dtype=torch.bfloat16) \
+        .contiguous(memory_format=torch.channels_last).requires_grad_(True)
+    x2 = x1.detach().clone().requires_grad_(True)
+    y1, y2 = conv(x1), ref(x2)
+    assert _rel_err(y1, y2) < 0.05
+    g = torch.randn_like(y1)
+    y1.backward(g)
+    y2.backward(g)
+    assert _rel_err(x1.grad, x2.grad) < 0.05
+    assert _rel_err(conv.weight.grad, ref.weight.grad) < 0.05
