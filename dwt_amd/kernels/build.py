@@ -10,8 +10,6 @@ from __future__ import annotations
 import glob
 import os
 import shutil
-import subprocess
-import sys
 
 HERE = os.path.dirname(os.path.abspath(__file__))
 SRC = os.path.join(HERE, "src")

@@ -9,8 +9,6 @@ the torch implementation (warned once).
 from __future__ import annotations
 
 import warnings
-from typing import Optional, Sequence
-
 import torch
 
 from . import dispatch

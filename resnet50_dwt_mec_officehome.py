@@ -11,7 +11,6 @@ import argparse
 import os
 
 import torch
-import torch.optim as optim
 from torch.utils.data import DataLoader, distributed as dist_data
 
 from dwt_amd.data import (Compose, ImageFolder, Lambda, Normalize, RandomCrop,

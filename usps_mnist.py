@@ -14,7 +14,6 @@ import argparse
 import os
 
 import torch
-import torch.optim as optim
 from torch.optim import lr_scheduler
 from torch.utils.data import DataLoader, distributed as dist_data
 
