@@ -1,0 +1,129 @@
+"""CPU fallback paths + remaining semantic corners."""
+import pytest
+import torch
+import torch.nn.functional as F
+
+from dwt_amd.ops import functional as Fdwt
+from dwt_amd.ops.mfma import MFMAConv2d, MFMALinear
+from dwt_amd.ops.optim import FusedAdam, FusedSGD
+from dwt_amd.ops.pooling import MaxPool2dDWT, global_avg_pool, max_pool2d
+from dwt_amd.ops.whitening import WTransform2d
+from dwt_amd.ops.batch_norm import DomainBatchNorm2d, DomainBatchNorm3d
+
+
+def test_mfma_layers_fall_back_on_cpu():
+    lin = MFMALinear(16, 8)
+    x = torch.randn(4, 16)
+    assert torch.allclose(lin(x), F.linear(x, lin.weight, lin.bias))
+    conv = MFMAConv2d(8, 8, 3, padding=1)
+    xc = torch.randn(2, 8, 6, 6)
+    ref = F.conv2d(xc, conv.weight, conv.bias, padding=1)
+    assert torch.allclose(conv(xc), ref)
+
+
+def test_pooling_falls_back_on_cpu():
+    x = torch.randn(2, 8, 10, 10)
+    assert torch.allclose(max_pool2d(x, 2, 2), F.max_pool2d(x, 2, 2))
+    assert torch.allclose(MaxPool2dDWT(3, 2, 1)(x),
+                          F.max_pool2d(x, 3, 2, 1))
+    assert torch.allclose(global_avg_pool(x),
+                          F.adaptive_avg_pool2d(x, (1, 1)).reshape(2, 8))
+
+
+def test_ce_loss_cpu():
+    x = torch.randn(7, 11, requires_grad=True)
+    t = torch.randint(0, 11, (7,))
+    loss = Fdwt.ce_loss(x, t)
+    ref = F.nll_loss(F.log_softmax(x.detach(), dim=1), t)
+    assert torch.allclose(loss, ref, atol=1e-6)
+    loss.backward()
+    assert x.grad is not None
+
+
+def test_fused_optimizers_fall_back_on_cpu():
+    torch.manual_seed(0)
+    m1 = torch.nn.Linear(5, 3)
+    m2 = torch.nn.Linear(5, 3)
+    m2.load_state_dict(m1.state_dict())
+    o1 = FusedSGD(m1.parameters(), lr=0.1, momentum=0.9, weight_decay=1e-3)
+    o2 = torch.optim.SGD(m2.parameters(), lr=0.1, momentum=0.9, weight_decay=1e-3)
+    for _ in range(3):
+        x = torch.randn(4, 5)
+        for m, o in ((m1, o1), (m2, o2)):
+            o.zero_grad()
+            m(x).pow(2).mean().backward()
+            o.step()
+    for p1, p2 in zip(m1.parameters(), m2.parameters()):
+        assert torch.allclose(p1, p2, atol=1e-7)
+
+    a1 = torch.nn.Linear(5, 3)
+    a2 = torch.nn.Linear(5, 3)
+    a2.load_state_dict(a1.state_dict())
+    oa1 = FusedAdam(a1.parameters(), lr=1e-3, weight_decay=1e-4)
+    oa2 = torch.optim.Adam(a2.parameters(), lr=1e-3, weight_decay=1e-4)
+    for _ in range(3):
+        x = torch.randn(4, 5)
+        for m, o in ((a1, oa1), (a2, oa2)):
+            o.zero_grad()
+            m(x).pow(2).mean().backward()
+            o.step()
+    for p1, p2 in zip(a1.parameters(), a2.parameters()):
+        assert torch.allclose(p1, p2, atol=1e-7)
+
+
+def test_whitening_track_running_stats_false():
+    """track=False: batch stats even in eval, buffers untouched
+    (whitening.py:42 guard is on track_running_stats)."""
+    mod = WTransform2d(8, 4, track_running_stats=False)
+    x = torch.randn(6, 8, 4, 4)
+    mod.train(); y_tr = mod(x)
+    mod.eval(); y_ev = mod(x)
+    assert torch.allclose(y_tr, y_ev, atol=1e-5)
+    assert torch.allclose(mod.running_mean, torch.zeros_like(mod.running_mean))
+
+
+def test_whitening_zca_module_mode():
+    mod = WTransform2d(8, 4, mode="zca")
+    x = torch.randn(16, 8, 5, 5) * 2 + 1
+    y = mod(x)
+    from dwt_amd.ops import oracle
+    cov_y = oracle.grouped_cov(y - oracle.channel_mean(y), 2)
+    eye = torch.eye(4).expand_as(cov_y)
+    assert (cov_y - eye).abs().max() < 0.1
+
+
+def test_bn3d():
+    ours = DomainBatchNorm3d(4, affine=False)
+    theirs = torch.nn.BatchNorm3d(4, affine=False)
+    x = torch.randn(3, 4, 2, 5, 5)
+    assert torch.allclose(ours(x), theirs(x), atol=1e-5)
+
+
+def test_bn_momentum_none_cumulative():
+    ours = DomainBatchNorm2d(4, affine=False)
+    ours.momentum = None
+    theirs = torch.nn.BatchNorm2d(4, affine=False, momentum=None)
+    for _ in range(3):
+        x = torch.randn(5, 4, 3, 3)
+        ours(x)
+        theirs(x)
+    assert torch.allclose(ours.running_mean, theirs.running_mean, atol=1e-6)
+    assert torch.allclose(ours.running_var, theirs.running_var, atol=1e-6)
+
+
+def test_usps_missing_file_message(tmp_path):
+    from dwt_amd.data import USPS
+    with pytest.raises(RuntimeError, match="no network"):
+        USPS(str(tmp_path), train=True)
+
+
+def test_add_relu_cpu():
+    a = torch.randn(3, 4, requires_grad=True)
+    b = torch.randn(3, 4, requires_grad=True)
+    out = Fdwt.add_relu(a, b)
+    ref = torch.relu(a.detach() + b.detach())
+    assert torch.allclose(out, ref)
+    out.sum().backward()
+    mask = (ref > 0).float()
+    assert torch.allclose(a.grad, mask)
+    assert torch.allclose(b.grad, mask)
