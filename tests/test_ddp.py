@@ -114,3 +114,20 @@ def test_bucket_assembly():
             seen.add(id(p))
     assert len(seen) == len(ddp.params)
     assert ddp.buckets[0].params[0] is ddp.params[-1]  # reverse order
+
+
+def test_digits_entrypoint_torchrun_cpu(tmp_path):
+    """Full digits entrypoint under torchrun ws=2 (gloo) — the DP path the
+    driver exercises on 8 GPUs, on CPU."""
+    import subprocess, sys, os as _os
+    repo = _os.path.dirname(_os.path.dirname(_os.path.abspath(__file__)))
+    r = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
+         "--master-port", "29712", "usps_mnist.py", "--synthetic",
+         "--synthetic_size", "64", "--epochs", "1", "--group_size", "4",
+         "--num_workers", "0", "--test_batch_size", "16",
+         "--log_interval", "1"],
+        cwd=repo, capture_output=True, text=True, timeout=900)
+    assert r.returncode == 0, (r.stdout[-800:], r.stderr[-1500:])
+    assert "Test set" in r.stdout
