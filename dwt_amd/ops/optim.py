@@ -134,6 +134,16 @@ class _FusedBase(Optimizer):
             self._fallback.load_state_dict(sd)
             return
         super().load_state_dict(sd)
+        # torch's load_state_dict casts floating state to the PARAM dtype;
+        # the fused kernels index state as fp32 — restore fp32 contiguous
+        # (a bf16-cast master/momentum would also corrupt the update).
+        for st in self.state.values():
+            for k, v in list(st.items()):
+                if torch.is_tensor(v) and v.is_floating_point() \
+                        and v.dtype != torch.float32:
+                    st[k] = v.float().contiguous()
+                elif torch.is_tensor(v) and not v.is_contiguous():
+                    st[k] = v.contiguous()
         self._tables = None
 
 
