@@ -20,7 +20,6 @@ import time
 
 import torch
 import torch.distributed as dist
-import torch.nn.functional as F
 
 
 def get_args():
