@@ -47,6 +47,8 @@ def train_infinite_collect_stats(args, model, device, source_train_loader,
 
     for i in range(start_iter, args.num_iters):
         model.train()
+        # scheduler stepped before the optimizer on purpose: reference
+        # behavior (resnet50_dwt_mec_officehome.py:403 — SURVEY quirk #9)
         exp_lr_scheduler.step()
         try:
             source_data, source_y = next(source_iter)

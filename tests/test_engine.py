@@ -116,3 +116,42 @@ def test_entrypoint_digits_mec_synthetic(capsys):
           "--group_size", "4", "--loss", "mec", "--num_workers", "0",
           "--test_batch_size", "16", "--log_interval", "1"])
     assert "Test set" in capsys.readouterr().out
+
+
+def test_jsonl_logger(tmp_path):
+    import json
+    from dwt_amd.engine.meters import JsonlLogger, AverageMeter, ThroughputMeter
+    path = str(tmp_path / "m.jsonl")
+    lg = JsonlLogger(path, rank=0)
+    lg.log(kind="train", step=1, loss=0.5)
+    lg.log(kind="test", acc=42.0)
+    lg.close()
+    rows = [json.loads(l) for l in open(path)]
+    assert rows[0]["kind"] == "train" and rows[1]["acc"] == 42.0
+    # rank != 0 writes nothing
+    lg2 = JsonlLogger(str(tmp_path / "n.jsonl"), rank=1)
+    lg2.log(kind="x")
+    lg2.close()
+    import os
+    assert not os.path.exists(str(tmp_path / "n.jsonl"))
+    m = AverageMeter(); m.update(2.0, 3); m.update(5.0)
+    assert abs(m.avg - 11.0 / 4) < 1e-9
+    tp = ThroughputMeter(); assert tp.tick(10) is None; assert tp.tick(10) > 0
+
+
+@pytest.mark.slow
+def test_officehome_sweep_single_pair(tmp_path):
+    import json
+    import subprocess, sys, os
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    out = str(tmp_path / "sweep.json")
+    r = subprocess.run(
+        [sys.executable, "benchmarks/officehome_sweep.py", "--synthetic",
+         "--synthetic_size", "24", "--num_iters", "1", "--out", out,
+         "--pairs", "Art:Clipart", "--source_batch_size", "4",
+         "--test_batch_size", "8", "--num_workers", "0", "--check_acc_step",
+         "100", "--log_interval", "1", "--stats_passes", "1"],
+        cwd=repo, capture_output=True, text=True, timeout=600)
+    assert r.returncode == 0, r.stderr[-1500:]
+    data = json.load(open(out))
+    assert "Art->Clipart" in data and "average" in data
