@@ -31,6 +31,7 @@ Math reproduced from the reference (not its code):
 """
 from __future__ import annotations
 
+import os
 from typing import List, Optional, Sequence
 
 import torch
@@ -117,6 +118,13 @@ def matfn_ns_backward(saved, w_bar: torch.Tensor):
 # ----------------------------------------------------------------------------
 
 
+def _dist_world():
+    import torch.distributed as dist
+    if dist.is_available() and dist.is_initialized() and dist.get_world_size() > 1:
+        return dist, dist.get_world_size()
+    return None, 1
+
+
 def _grouped_apply(xn: torch.Tensor, w: torch.Tensor, num_groups: int) -> torch.Tensor:
     """y[:, Gg+i] = sum_j W[G, i, j] xn[:, Gg+j] — the grouped 1x1 'conv'."""
     n, c, h, hw = xn.shape
@@ -174,12 +182,30 @@ class WhitenMulti(torch.autograd.Function):
         wmats = []
         saved_mat = []
         use_batch = training or not track
+        stats_sync = cfg.get("stats_sync", False) and use_batch
+        dist, world = _dist_world() if stats_sync else (None, 1)
+        ctx.world = world
         for p in range(parts):
             xp = x[p * b:(p + 1) * b].to(comp_dtype)
             if use_batch:
-                m = xp.mean(dim=(0, 2, 3)).view(1, c, 1, 1)
-                xn = xp - m
-                cov = oracle.grouped_cov(xn, num_groups)
+                if dist is not None:
+                    # cross-rank batch statistics (SyncBN-style): all-reduce
+                    # the raw sums and co-moments, stats over the GLOBAL batch
+                    m_local = xp.sum(dim=(0, 2, 3))
+                    t = xp.permute(1, 0, 2, 3).reshape(num_groups, g, -1)
+                    pm = torch.bmm(t, t.transpose(1, 2))
+                    pack = torch.cat([m_local, pm.reshape(-1)])
+                    dist.all_reduce(pack)
+                    m_cnt = b * h * w * world
+                    m = (pack[:c] / m_cnt).view(1, c, 1, 1)
+                    exx = pack[c:].reshape(num_groups, g, g) / m_cnt
+                    mg = m.reshape(num_groups, g, 1)
+                    cov = exx - torch.bmm(mg, mg.transpose(1, 2))
+                    xn = xp - m
+                else:
+                    m = xp.mean(dim=(0, 2, 3)).view(1, c, 1, 1)
+                    xn = xp - m
+                    cov = oracle.grouped_cov(xn, num_groups)
                 cov_s = oracle.shrink_cov(cov, eps)
             else:
                 m = running_means[p].reshape(1, c, 1, 1).to(comp_dtype)
@@ -249,7 +275,12 @@ class WhitenMulti(torch.autograd.Function):
             dy0 = dy.to(comp_dtype)
 
         dx = torch.empty_like(x)
-        m_count = b * h * w
+        world = getattr(ctx, "world", 1)
+        stats_sync = cfg.get("stats_sync", False) and use_batch and world > 1
+        dist = None
+        if stats_sync:
+            import torch.distributed as dist
+        m_count = b * h * w * world
         for p in range(parts):
             sl = slice(p * b, (p + 1) * b)
             xp = x[sl].to(comp_dtype)
@@ -262,14 +293,20 @@ class WhitenMulti(torch.autograd.Function):
             if use_batch:
                 # through the whitening matrix and the covariance
                 w_bar = _grouped_outer(dyp, xn, num_groups)
+                if stats_sync:
+                    dist.all_reduce(w_bar)
                 if mode == "chol":
                     a_bar = matfn_chol_backward(wmat, ctx.saved_mat[p], w_bar)
                 else:
                     a_bar = matfn_ns_backward(ctx.saved_mat[p], w_bar)
                 s_mat = (1.0 - eps) / m_count * (a_bar + a_bar.transpose(-1, -2))
                 dt = dt + _grouped_apply(xn, s_mat, num_groups)
-                # through the mean (m = per-channel mean of x)
-                dt = dt - dt.mean(dim=(0, 2, 3), keepdim=True)
+                # through the mean (m = per-channel mean of ALL ranks' x)
+                corr = dt.mean(dim=(0, 2, 3), keepdim=True)
+                if stats_sync:
+                    dist.all_reduce(corr)
+                    corr /= world
+                dt = dt - corr
             dx[sl] = dt.to(x.dtype)
 
         return dx, dgamma, dbeta, None, None, None
@@ -295,6 +332,12 @@ def whiten_multi(
     cfg = dict(parts=parts, num_groups=num_groups, eps=eps, momentum=momentum,
                training=training, mode=mode, relu=relu, ns_iters=ns_iters,
                track_running_stats=track_running_stats)
+    if os.environ.get("DWT_AMD_STATS_SYNC") == "1":
+        # cross-rank batch statistics ('sync' stats mode) —
+        # currently served by the torch path (exact; HIP fast path is
+        # per-rank stats + parallel.sync_stats())
+        cfg["stats_sync"] = True
+        return WhitenMulti.apply(x, gamma, beta, running_means, running_vars, cfg)
     if x.is_cuda:
         from ..kernels import dispatch
         if dispatch.available():
@@ -335,13 +378,24 @@ class BatchNormMulti(torch.autograd.Function):
         xhat = torch.empty_like(x)
         means, invstds = [], []
         red_dims = (0, 2, 3) if spatial else (0,)
+        stats_sync = cfg.get("stats_sync", False) and use_batch
+        dist, world = _dist_world() if stats_sync else (None, 1)
+        ctx.world = world
         for p in range(parts):
             sl = slice(p * b, (p + 1) * b)
             xp = x[sl].to(comp_dtype)
             if use_batch:
-                m = xp.mean(dim=red_dims)
-                var = xp.var(dim=red_dims, unbiased=False)
-                cnt = xp.numel() // c
+                if dist is not None:
+                    pack = torch.stack([xp.sum(dim=red_dims),
+                                        (xp * xp).sum(dim=red_dims)])
+                    dist.all_reduce(pack)
+                    cnt = (xp.numel() // c) * world
+                    m = pack[0] / cnt
+                    var = (pack[1] / cnt - m * m).clamp_min(0)
+                else:
+                    m = xp.mean(dim=red_dims)
+                    var = xp.var(dim=red_dims, unbiased=False)
+                    cnt = xp.numel() // c
                 if training and track and running_means is not None:
                     with torch.no_grad():
                         rm, rv = running_means[p], running_vars[p]
@@ -405,6 +459,12 @@ class BatchNormMulti(torch.autograd.Function):
             if use_batch:
                 mean_dxh = dxh.mean(dim=red_dims, keepdim=True)
                 mean_dxh_xh = (dxh * xh).mean(dim=red_dims, keepdim=True)
+                world = getattr(ctx, "world", 1)
+                if cfg.get("stats_sync", False) and world > 1:
+                    import torch.distributed as dist
+                    pack = torch.stack([mean_dxh, mean_dxh_xh])
+                    dist.all_reduce(pack)
+                    mean_dxh, mean_dxh_xh = pack[0] / world, pack[1] / world
                 dxp = (dxh - mean_dxh - xh * mean_dxh_xh) * istd
             else:
                 dxp = dxh * istd
@@ -428,6 +488,9 @@ def batch_norm_multi(
 ) -> torch.Tensor:
     cfg = dict(parts=parts, eps=eps, momentum=momentum, training=training,
                relu=relu, track_running_stats=track_running_stats)
+    if os.environ.get("DWT_AMD_STATS_SYNC") == "1":
+        cfg["stats_sync"] = True
+        return BatchNormMulti.apply(x, gamma, beta, running_means, running_vars, cfg)
     if x.is_cuda:
         from ..kernels import dispatch
         if dispatch.available():

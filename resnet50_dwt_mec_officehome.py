@@ -60,6 +60,9 @@ def build_args(argv=None):
     parser.add_argument('--checkpoint_every', type=int, default=1000)
     parser.add_argument('--resume', action='store_true')
     parser.add_argument('--metrics_jsonl', type=str, default='')
+    parser.add_argument('--stats_mode', choices=['local', 'sync'], default='local',
+                        help="per-rank norm statistics (stock-DDP-like) or "
+                             "cross-rank synced batch stats")
     parser.add_argument('--stats_passes', type=int, default=10,
                         help='target-stats re-estimation passes before final test')
     return parser.parse_args(argv)
@@ -113,6 +116,8 @@ def build_loaders(args, rank, world):
 
 def main(argv=None):
     args = build_args(argv)
+    if getattr(args, 'stats_mode', 'local') == 'sync':
+        os.environ['DWT_AMD_STATS_SYNC'] = '1'
     rank, world, local_rank = init_distributed_from_env()
     seed_everything(args.seed, rank)
     device = torch.device(f'cuda:{local_rank}' if torch.cuda.is_available() else 'cpu')

@@ -131,3 +131,62 @@ def test_digits_entrypoint_torchrun_cpu(tmp_path):
         cwd=repo, capture_output=True, text=True, timeout=900)
     assert r.returncode == 0, (r.stdout[-800:], r.stderr[-1500:])
     assert "Test set" in r.stdout
+
+
+def _run_stats_sync_parity(rank, world, port, results):
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    try:
+        from dwt_amd.ops.functional import WhitenMulti, BatchNormMulti
+        torch.manual_seed(0)
+        full = torch.randn(8, 8, 5, 5, dtype=torch.float64)
+        local = full[rank * 4:(rank + 1) * 4].clone().requires_grad_(True)
+        ref = full.clone().requires_grad_(True)
+
+        # whitening: synced 2-rank stats == single-process full-batch stats
+        cfg = dict(parts=1, num_groups=2, eps=1e-3, momentum=0.1,
+                   training=True, mode="chol", relu=False, stats_sync=True)
+        out = WhitenMulti.apply(local, None, None, None, None, cfg)
+        cfg1 = dict(cfg, stats_sync=False)
+        out_ref = WhitenMulti.apply(ref, None, None, None, None, cfg1)
+        assert torch.allclose(out, out_ref[rank * 4:(rank + 1) * 4], atol=1e-9)
+        out.sum().backward()
+        out_ref.sum().backward()
+        assert torch.allclose(local.grad, ref.grad[rank * 4:(rank + 1) * 4],
+                              atol=1e-9), (local.grad - ref.grad[rank*4:(rank+1)*4]).abs().max()
+
+        # BN: same property
+        local2 = full[rank * 4:(rank + 1) * 4].clone().requires_grad_(True)
+        ref2 = full.clone().requires_grad_(True)
+        bcfg = dict(parts=1, eps=1e-5, momentum=0.1, training=True,
+                    relu=True, stats_sync=True)
+        o2 = BatchNormMulti.apply(local2, None, None, None, None, bcfg)
+        o2r = BatchNormMulti.apply(ref2, None, None, None, None,
+                                   dict(bcfg, stats_sync=False))
+        assert torch.allclose(o2, o2r[rank * 4:(rank + 1) * 4], atol=1e-9)
+        o2.sum().backward()
+        o2r.sum().backward()
+        assert torch.allclose(local2.grad, ref2.grad[rank * 4:(rank + 1) * 4],
+                              atol=1e-9)
+        if rank == 0:
+            results.put("ok")
+    finally:
+        dist.destroy_process_group()
+
+
+def test_stats_sync_matches_full_batch():
+    """stats_sync=True on 2 ranks == single-process stats over the
+    concatenated batch (forward AND backward) — the 'sync' stats mode of
+    SURVEY §2.3."""
+    port = 29871
+    ctx = mp.get_context("spawn")
+    results = ctx.Queue()
+    procs = [ctx.Process(target=_run_stats_sync_parity, args=(r, 2, port, results))
+             for r in range(2)]
+    for p in procs:
+        p.start()
+    for p in procs:
+        p.join(timeout=180)
+    assert all(p.exitcode == 0 for p in procs), [p.exitcode for p in procs]
+    assert results.get(timeout=5) == "ok"

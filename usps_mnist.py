@@ -57,6 +57,9 @@ def build_args(argv=None):
     parser.add_argument('--checkpoint_path', type=str, default='')
     parser.add_argument('--resume', action='store_true')
     parser.add_argument('--metrics_jsonl', type=str, default='')
+    parser.add_argument('--stats_mode', choices=['local', 'sync'], default='local',
+                        help="per-rank norm statistics (stock-DDP-like) or "
+                             "cross-rank synced batch stats")
     return parser.parse_args(argv)
 
 
@@ -99,6 +102,8 @@ def build_datasets(args):
 
 def main(argv=None):
     args = build_args(argv)
+    if getattr(args, 'stats_mode', 'local') == 'sync':
+        os.environ['DWT_AMD_STATS_SYNC'] = '1'
     assert args.source != args.target, "source and target datasets can not be the same"
     rank, world, local_rank = init_distributed_from_env()
     seed_everything(args.seed, rank)
