@@ -26,13 +26,22 @@ def _whiten_layout(x, c):
     return "nchw"
 
 
+def _bn_cl_supported(c):
+    """The NHWC BN kernels map lanes as t %% NCH with NCH = min(C,1024)/4;
+    rows advance by blockDim/NCH — NCH must divide the block (else boundary
+    rows are double-counted)."""
+    if c % 4 != 0 or (c > 1024 and c % 1024 != 0):
+        return False
+    nch = min(c, 1024) // 4
+    return 256 % nch == 0
+
+
 def _bn_layout(x, c):
-    if x.dim() == 2 and c % 4 == 0 and x.is_contiguous():
+    if x.dim() == 2 and x.is_contiguous() and _bn_cl_supported(c):
         return "cl"  # (N, C) IS channels-last: C contiguous per position
     if x.dim() == 4 and x.is_contiguous(memory_format=torch.channels_last) \
-            and not x.is_contiguous():
-        if c % 4 == 0 and (c <= 1024 or c % 1024 == 0):
-            return "cl"
+            and not x.is_contiguous() and _bn_cl_supported(c):
+        return "cl"
     return "nchw"
 
 
