@@ -17,12 +17,15 @@ _FAST_G = (2, 4, 8)
 _warned = set()
 
 
-def _whiten_layout(x, c):
-    """'cl' when x is channels_last and the NHWC kernels support C."""
+def _whiten_layout(x, c, g):
+    """'cl' when x is channels_last and the NHWC kernels support (C, g):
+    the lane map needs GW = min(C,256)/g to divide the 256-thread block."""
     if x.dim() == 4 and x.is_contiguous(memory_format=torch.channels_last) \
             and not x.is_contiguous():
         if c % 64 == 0 and (c <= 256 or c % 256 == 0):
-            return "cl"
+            gw = min(c, 256) // g
+            if gw > 0 and 256 % gw == 0:
+                return "cl"
     return "nchw"
 
 
@@ -72,7 +75,7 @@ class _HipWhitenMulti(torch.autograd.Function):
         use_batch = training or not track
 
         c = x.shape[1]
-        layout = _whiten_layout(x, c)
+        layout = _whiten_layout(x, c, g)
         if layout == "nchw":
             x = x.contiguous()
         n, c, h, w = x.shape
