@@ -22,7 +22,9 @@ def _whiten_layout(x, c, g):
     the lane map needs GW = min(C,256)/g to divide the 256-thread block."""
     if x.dim() == 4 and x.is_contiguous(memory_format=torch.channels_last) \
             and not x.is_contiguous():
-        if c % 64 == 0 and (c <= 256 or c % 256 == 0):
+        if c % 64 == 0 and (c <= 256 or c % 256 == 0) and g in (2, 4):
+            # g=8 NHWC instantiations spill registers (profiles/
+            # kernel_resources.md) — route them to the NCHW path
             gw = min(c, 256) // g
             if gw > 0 and 256 % gw == 0:
                 return "cl"

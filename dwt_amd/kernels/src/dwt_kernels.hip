@@ -1808,7 +1808,7 @@ void whiten_stats(Tensor x, Tensor acc, Tensor mean, Tensor cov, int64_t g) {
   const int n_groups = C / g;
   DISPATCH_FT(x, "whiten_stats", [&] {
     constexpr int VW = dwt::VecTraits<scalar_t>::W;
-    const bool vec = can_vectorize<scalar_t>(x, HW) && (M % VW == 0);
+    const bool vec = g <= 4 && can_vectorize<scalar_t>(x, HW) && (M % VW == 0);
     const int threads = 256;
     dim3 grid(reduce_blocks(M, threads, vec ? VW : 1, n_groups), n_groups);
     auto launch = [&](auto gconst, auto vconst) {
@@ -2102,7 +2102,9 @@ void whiten_bwd_reduce(Tensor x, Tensor dout, Tensor out, Tensor mean, Tensor W,
   const int n_groups = C / g;
   DISPATCH_FT(x, "whiten_bwd_reduce", [&] {
     constexpr int VW = dwt::VecTraits<scalar_t>::W;
-    const bool vec = can_vectorize<scalar_t>(x, HW) &&
+    // g==8 vectorized variants spill (G*G + 3*G*VW live floats) — scalar
+    // loop for them (profiles/kernel_resources.md)
+    const bool vec = g <= 4 && can_vectorize<scalar_t>(x, HW) &&
                      can_vectorize<scalar_t>(dout, HW) && (M % VW == 0);
     const int threads = 256;
     auto launch = [&](auto gconst, auto vconst) {
@@ -2138,7 +2140,7 @@ void whiten_bwd_apply(Tensor x, Tensor dout, Tensor out, Tensor mean, Tensor W,
   const int n_groups = C / g;
   DISPATCH_FT(x, "whiten_bwd_apply", [&] {
     constexpr int VW = dwt::VecTraits<scalar_t>::W;
-    const bool vec = can_vectorize<scalar_t>(x, HW) &&
+    const bool vec = g <= 4 && can_vectorize<scalar_t>(x, HW) &&
                      can_vectorize<scalar_t>(dx, HW);
     const int threads = 256;
     auto launch = [&](auto gconst, auto vconst) {
