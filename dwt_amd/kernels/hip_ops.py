@@ -81,6 +81,7 @@ class _HipWhitenMulti(torch.autograd.Function):
         if layout == "nchw":
             x = x.contiguous()
         n, c, h, w = x.shape
+        assert n % parts == 0,             f"batch {n} not divisible by {parts} domain branches"
         b = n // parts
         n_groups = c // g
         dev = x.device
@@ -247,6 +248,7 @@ class _HipBatchNormMulti(torch.autograd.Function):
         if layout == "nchw" or x.dim() == 2:
             x = x.contiguous()
         n = x.shape[0]
+        assert n % parts == 0,             f"batch {n} not divisible by {parts} domain branches"
         b = n // parts
         dev = x.device
         cnt = (x.numel() // parts) // c
