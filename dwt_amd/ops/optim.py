@@ -126,13 +126,20 @@ class _FusedBase(Optimizer):
         self._ensure_mode()
         if not self._use_hip:
             return self._fallback.state_dict()
-        return super().state_dict()
+        sd = super().state_dict()
+        if hasattr(self, "_step_count"):
+            sd["fused_step_count"] = self._step_count
+        return sd
 
     def load_state_dict(self, sd):
         self._ensure_mode()
         if not self._use_hip:
             self._fallback.load_state_dict(sd)
             return
+        sd = dict(sd)
+        step_count = sd.pop("fused_step_count", None)
+        if step_count is not None and hasattr(self, "_step_count"):
+            self._step_count = step_count
         super().load_state_dict(sd)
         # torch's load_state_dict casts floating state to the PARAM dtype;
         # the fused kernels index state as fp32 — restore fp32 contiguous
