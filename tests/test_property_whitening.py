@@ -27,7 +27,7 @@ def whiten_case(draw):
     return parts, b, c, groups, h, w, mode, seed
 
 
-@settings(max_examples=25, deadline=None)
+@settings(max_examples=25, deadline=None, derandomize=True)
 @given(whiten_case())
 def test_whiten_forward_backward_property(case):
     parts, b, c, groups, h, w, mode, seed = case
@@ -65,7 +65,7 @@ def test_whiten_forward_backward_property(case):
     assert torch.allclose(gx, gx2, atol=1e-7), (gx - gx2).abs().max()
 
 
-@settings(max_examples=15, deadline=None)
+@settings(max_examples=15, deadline=None, derandomize=True)
 @given(st.integers(0, 2 ** 16), st.sampled_from([2, 4, 8]),
        st.sampled_from(["chol", "zca"]))
 def test_whitening_decorrelates_property(seed, g, mode):
