@@ -155,3 +155,22 @@ def test_officehome_sweep_single_pair(tmp_path):
     assert r.returncode == 0, r.stderr[-1500:]
     data = json.load(open(out))
     assert "Art->Clipart" in data and "average" in data
+
+
+def test_officehome_resume_fast_forwards_scheduler():
+    """start_iter fast-forwards the MultiStepLR past its milestone."""
+    from dwt_amd.ops.optim import FusedSGD
+    src = SyntheticOfficeHome(8, num_classes=5, img_size=64, seed=1)
+    tgt = SyntheticOfficeHome(8, num_classes=5, img_size=64, transform_aug=True, seed=2)
+    tst = SyntheticOfficeHome(4, num_classes=5, img_size=64, seed=2)
+    src_loader = DataLoader(src, batch_size=4, drop_last=True)
+    tgt_loader = DataLoader(tgt, batch_size=4, drop_last=True)
+    test_loader = DataLoader(tst, batch_size=4)
+    model = ResNetDWT(Bottleneck, [1, 1, 1, 1], None, num_classes=5)
+    opt = FusedSGD(model.parameters(), lr=1.0, momentum=0.0)
+    args = _args(num_iters=6002, check_acc_step=10**9, stats_passes=0)
+    # start past the 6000-iter milestone: lr must arrive decayed by 0.1
+    train_infinite_collect_stats(args, model, torch.device("cpu"), src_loader,
+                                 tgt_loader, opt, 0.1, test_loader,
+                                 start_iter=6001, stats_passes=0)
+    assert abs(opt.param_groups[0]["lr"] - 0.1) < 1e-9
