@@ -199,6 +199,51 @@ DEV_INLINE void load_b(const bf16* __restrict__ wgt, int ncols, int K,
   }
 }
 
+// wgrad B-operand gather (see mfma_conv2d_wgrad): row n' = (r*KW+s)*Cin+ci,
+// k = flattened (n, p, q) output position of the forward conv; per-element
+// decode (k-chunks cross q rows), zeros outside bounds or past the real NPQ
+// (K is padded to a multiple of 8 for the dense A loads).
+DEV_INLINE void load_b_wgrad(const bf16* __restrict__ x, const ConvParams& cp,
+                             int n0, int k0, StageRegs& rg) {
+  const int t = threadIdx.x;
+  const int npq = cp.Cdy;  // real (unpadded) N*P*Q
+#pragma unroll
+  for (int c = 0; c < 4; ++c) {
+    const int idx = t + c * THREADS;
+    const int row = idx / (BK / 8);
+    const int kc = (idx % (BK / 8)) * 8;
+    const int nrow = n0 + row;
+    const int kg = k0 + kc;
+    if (nrow >= cp.Cout || kg >= cp.K) {  // cp.Cout = R*S*Cin columns of dw
+      rg.c[c] = make_uint4(0, 0, 0, 0);
+      continue;
+    }
+    const int ci = nrow % cp.Cin;
+    const int rs = nrow / cp.Cin;
+    const int sx = rs % cp.KW;
+    const int r = rs / cp.KW;
+    unsigned short tmp[8];
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      const int k = kg + j;
+      unsigned short v = 0;
+      if (k < npq) {
+        const int q = k % cp.Q;
+        const int np = k / cp.Q;
+        const int pp = np % cp.P;
+        const int n = np / cp.P;
+        const int h = pp * cp.stride - cp.pad + r;
+        const int w = q * cp.stride - cp.pad + sx;
+        if (h >= 0 && h < cp.H && w >= 0 && w < cp.W)
+          v = *reinterpret_cast<const unsigned short*>(
+              x + (((int64_t)n * cp.H + h) * cp.W + w) * cp.Cin + ci);
+      }
+      tmp[j] = v;
+    }
+    rg.c[c] = *reinterpret_cast<const uint4*>(tmp);
+  }
+}
+
 DEV_INLINE void write_tile(const StageRegs& rg, unsigned short* lds) {
   const int t = threadIdx.x;
 #pragma unroll
@@ -215,7 +260,8 @@ DEV_INLINE void write_tile(const StageRegs& rg, unsigned short* lds) {
 // single-buffer loop wins for short-K and for the implicit gather path
 // (within-shape A/B on the R50 shapes).  SWZ: XCD-aware bijective block
 // remap (guide T1) — only when the grid has several N-tiles to share.
-template <int MODE, bool RELU, bool HAS_BIAS, bool PIPE, bool SWZ>
+template <int MODE, bool RELU, bool HAS_BIAS, bool PIPE, bool SWZ,
+          bool WGRAD_B = false>
 __global__ __launch_bounds__(THREADS, 2) void conv_implicit_gemm_kernel(
     const bf16* __restrict__ x, const bf16* __restrict__ wgt,
     const float* __restrict__ bias, bf16* __restrict__ out, ConvParams cp,
@@ -252,7 +298,8 @@ __global__ __launch_bounds__(THREADS, 2) void conv_implicit_gemm_kernel(
   StageRegs ra, rb;
   if (PIPE) {
     load_a<MODE>(x, cp, m0, 0, ra);
-    load_b(wgt, cp.Cout, cp.K, n0, 0, rb);
+    if (WGRAD_B) load_b_wgrad(wgt, cp, n0, 0, rb);
+    else load_b(wgt, cp.Cout, cp.K, n0, 0, rb);
     write_tile(ra, lds_a[0]);
     write_tile(rb, lds_b[0]);
     __syncthreads();
@@ -265,11 +312,13 @@ __global__ __launch_bounds__(THREADS, 2) void conv_implicit_gemm_kernel(
       // MFMA phase and are only waited for at the ds_write below
       if (t + 1 < nk) {
         load_a<MODE>(x, cp, m0, (t + 1) * BK, ra);
-        load_b(wgt, cp.Cout, cp.K, n0, (t + 1) * BK, rb);
+        if (WGRAD_B) load_b_wgrad(wgt, cp, n0, (t + 1) * BK, rb);
+        else load_b(wgt, cp.Cout, cp.K, n0, (t + 1) * BK, rb);
       }
     } else {
       load_a<MODE>(x, cp, m0, t * BK, ra);
-      load_b(wgt, cp.Cout, cp.K, n0, t * BK, rb);
+      if (WGRAD_B) load_b_wgrad(wgt, cp, n0, t * BK, rb);
+      else load_b(wgt, cp.Cout, cp.K, n0, t * BK, rb);
       write_tile(ra, lds_a[0]);
       write_tile(rb, lds_b[0]);
       __syncthreads();
@@ -448,4 +497,32 @@ void mfma_conv2d_dgrad(Tensor dy, Tensor wd, Tensor dx, int64_t N, int64_t H,
       dim3(mtiles * ntiles), dim3(dwtmm::THREADS), 0, dwtmm_stream(),
       (const c10::BFloat16*)dy.data_ptr(), (const c10::BFloat16*)wd.data_ptr(),
       nullptr, (c10::BFloat16*)dx.data_ptr(), cp, mtiles, ntiles);
+}
+
+
+// wgrad: dw[co][(r,s,ci)] (= channels_last weight storage) as the GEMM
+//   A[co][k=npq] = dy^T (host-transposed, K padded to %8)
+//   Bt[(r,s,ci)][k] = x patches (gathered in-kernel)
+// cp.M = Cout, cp.Cout = KH*KW*Cin (dw columns), cp.K = padded NPQ,
+// cp.Cdy = real NPQ.
+void mfma_conv2d_wgrad(Tensor dyT, Tensor x, Tensor dw, int64_t N, int64_t H,
+                       int64_t W, int64_t Cin, int64_t P, int64_t Q,
+                       int64_t Cout, int64_t KH, int64_t KW, int64_t stride,
+                       int64_t pad) {
+  dwtmm::ConvParams cp{};
+  cp.N = N; cp.H = H; cp.W = W; cp.Cin = Cin;
+  cp.P = P; cp.Q = Q;
+  cp.KH = KH; cp.KW = KW; cp.stride = stride; cp.pad = pad;
+  cp.M = Cout;
+  cp.Cout = KH * KW * Cin;
+  cp.K = dyT.size(1);          // padded
+  cp.Cdy = N * P * Q;          // real
+  const int mtiles = (cp.M + dwtmm::BM - 1) / dwtmm::BM;
+  const int ntiles = (cp.Cout + dwtmm::BN - 1) / dwtmm::BN;
+  hipLaunchKernelGGL(
+      (dwtmm::conv_implicit_gemm_kernel<dwtmm::A_DENSE, false, false, false,
+                                        false, true>),
+      dim3(mtiles * ntiles), dim3(dwtmm::THREADS), 0, dwtmm_stream(),
+      (const c10::BFloat16*)dyT.data_ptr(), (const c10::BFloat16*)x.data_ptr(),
+      nullptr, (c10::BFloat16*)dw.data_ptr(), cp, mtiles, ntiles);
 }

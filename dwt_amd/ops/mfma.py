@@ -165,3 +165,49 @@ class MFMAConv2d(nn.Conv2d):
                 return _MFMAConv2dFn.apply(x, self.weight, self.bias,
                                            self.stride[0], self.padding[0])
         return super().forward(x)
+
+
+def _wgrad_patches_reference(x_nhwc, shape, stride, padding, npq_pad=None,
+                             dtype=torch.float64):
+    """CPU/torch reference of the kernel's wgrad B-operand gather: returns
+    Bt [(r,s,ci), k=npq] built with EXACTLY the kernel's index decode
+    (mfma_conv.hip::load_b_wgrad) — used by the index-math test."""
+    n, h, w, cin = x_nhwc.shape
+    cout, _, kh, kw = shape
+    p = (h + 2 * padding - kh) // stride + 1
+    q = (w + 2 * padding - kw) // stride + 1
+    npq = n * p * q
+    k_tot = npq_pad or npq
+    cols = torch.zeros(kh * kw * cin, k_tot, dtype=dtype)
+    ks = torch.arange(npq)
+    qq = ks % q
+    pp = (ks // q) % p
+    nn = ks // (q * p)
+    for r in range(kh):
+        for s in range(kw):
+            hh = pp * stride - padding + r
+            ww = qq * stride - padding + s
+            ok = (hh >= 0) & (hh < h) & (ww >= 0) & (ww < w)
+            vals = torch.zeros(npq, cin, dtype=dtype)
+            vals[ok] = x_nhwc[nn[ok], hh[ok].clamp(0), ww[ok].clamp(0)].to(dtype)
+            for ci in range(cin):
+                cols[(r * kw + s) * cin + ci, :npq] = vals[:, ci]
+    return cols
+
+
+def conv2d_wgrad(dy: torch.Tensor, x: torch.Tensor, weight_shape,
+                 stride: int = 1, padding: int = 0) -> torch.Tensor:
+    """dw for an NHWC conv on the MFMA kernel: A = dy^T (host transpose),
+    B = input patches gathered in-kernel; output IS the channels_last
+    weight storage [co][r][s][ci]."""
+    cout, cin, kh, kw = weight_shape
+    n, _, p, q = dy.shape
+    dyc = dy.contiguous(memory_format=torch.channels_last)
+    dyt = dyc.permute(0, 2, 3, 1).reshape(n * p * q, cout).t().contiguous()
+    dyt = _pad_k(dyt)
+    dw = torch.empty(weight_shape, device=dy.device, dtype=torch.bfloat16,
+                     memory_format=torch.channels_last)
+    xc = x.contiguous(memory_format=torch.channels_last)
+    _ext().mfma_conv2d_wgrad(dyt, xc, dw, n, x.shape[2], x.shape[3], cin,
+                             p, q, cout, kh, kw, stride, padding)
+    return dw
