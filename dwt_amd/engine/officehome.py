@@ -45,7 +45,36 @@ def train_infinite_collect_stats(args, model, device, source_train_loader,
     dtype = next(model.parameters()).dtype
     tp = ThroughputMeter()
 
+    progress = {"iter": start_iter}
+    try:
+        return _train_loop(args, model, device, source_iter, target_iter,
+                           source_train_loader, target_train_loader,
+                           optimizer, exp_lr_scheduler, lambda_mec_loss,
+                           target_test_loader, logger, grad_sync, start_iter,
+                           checkpoint_path, checkpoint_every, stats_passes,
+                           dtype, tp, progress)
+    except Exception:
+        # failure handling: persist an emergency checkpoint so a
+        # torchrun restart can --resume instead of losing the run
+        if checkpoint_path:
+            try:
+                ckpt.save_training_state(checkpoint_path + ".emergency", model,
+                                         optimizer, exp_lr_scheduler,
+                                         iteration=progress["iter"])
+                print(f"emergency checkpoint written: {checkpoint_path}.emergency")
+            except Exception as exc:  # keep the original error primary
+                print(f"emergency checkpoint failed: {exc}")
+        raise
+
+
+def _train_loop(args, model, device, source_iter, target_iter,
+                source_train_loader, target_train_loader, optimizer,
+                exp_lr_scheduler, lambda_mec_loss, target_test_loader, logger,
+                grad_sync, start_iter, checkpoint_path, checkpoint_every,
+                stats_passes, dtype, tp, progress=None):
     for i in range(start_iter, args.num_iters):
+        if progress is not None:
+            progress["iter"] = i
         model.train()
         # scheduler stepped before the optimizer on purpose: reference
         # behavior (resnet50_dwt_mec_officehome.py:403 — SURVEY quirk #9)

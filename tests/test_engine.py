@@ -174,3 +174,41 @@ def test_officehome_resume_fast_forwards_scheduler():
                                  tgt_loader, opt, 0.1, test_loader,
                                  start_iter=6001, stats_passes=0)
     assert abs(opt.param_groups[0]["lr"] - 0.1) < 1e-9
+
+
+def test_emergency_checkpoint_on_failure(tmp_path):
+    from dwt_amd.models import checkpoint as ckptmod
+    src = SyntheticOfficeHome(8, num_classes=5, img_size=64, seed=1)
+    tgt = SyntheticOfficeHome(8, num_classes=5, img_size=64, transform_aug=True, seed=2)
+    tst = SyntheticOfficeHome(4, num_classes=5, img_size=64, seed=2)
+    src_loader = DataLoader(src, batch_size=4, drop_last=True)
+    tgt_loader = DataLoader(tgt, batch_size=4, drop_last=True)
+    test_loader = DataLoader(tst, batch_size=4)
+    model = ResNetDWT(Bottleneck, [1, 1, 1, 1], None, num_classes=5)
+    opt = torch.optim.SGD(model.parameters(), lr=1e-3)
+    path = str(tmp_path / "oh.pt")
+
+    class Boom(RuntimeError):
+        pass
+
+    calls = {"n": 0}
+    orig_forward = model.forward
+
+    def exploding(x):
+        calls["n"] += 1
+        if calls["n"] >= 3:
+            raise Boom("injected fault")
+        return orig_forward(x)
+
+    model.forward = exploding
+    args = _args(num_iters=10, check_acc_step=10**9)
+    with pytest.raises(Boom):
+        train_infinite_collect_stats(args, model, torch.device("cpu"),
+                                     src_loader, tgt_loader, opt, 0.1,
+                                     test_loader, checkpoint_path=path)
+    import os
+    assert os.path.exists(path + ".emergency")
+    model.forward = orig_forward
+    model2 = ResNetDWT(Bottleneck, [1, 1, 1, 1], None, num_classes=5)
+    it = ckptmod.load_training_state(path + ".emergency", model2)
+    assert it == 2  # failed during iteration 2's forward
