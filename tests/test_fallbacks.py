@@ -127,3 +127,37 @@ def test_add_relu_cpu():
     mask = (ref > 0).float()
     assert torch.allclose(a.grad, mask)
     assert torch.allclose(b.grad, mask)
+
+
+def test_fused_optimizer_state_roundtrip_cpu():
+    torch.manual_seed(0)
+    m1 = torch.nn.Linear(6, 4)
+    o1 = FusedSGD(m1.parameters(), lr=0.1, momentum=0.9)
+    for _ in range(2):
+        o1.zero_grad()
+        m1(torch.randn(3, 6)).pow(2).mean().backward()
+        o1.step()
+    sd = o1.state_dict()
+
+    m2 = torch.nn.Linear(6, 4)
+    m2.load_state_dict(m1.state_dict())
+    o2 = FusedSGD(m2.parameters(), lr=0.1, momentum=0.9)
+    o2.load_state_dict(sd)
+    x = torch.randn(3, 6)
+    for m, o in ((m1, o1), (m2, o2)):
+        o.zero_grad()
+        m(x).pow(2).mean().backward()
+        o.step()
+    for p1, p2 in zip(m1.parameters(), m2.parameters()):
+        assert torch.allclose(p1, p2, atol=1e-7)
+
+
+def test_mfma_gemm_pad_helper():
+    from dwt_amd.ops.mfma import _pad_k
+    t = torch.randn(3, 13)
+    p = _pad_k(t)
+    assert p.shape == (3, 16)
+    assert torch.allclose(p[:, :13], t)
+    assert p[:, 13:].abs().sum() == 0
+    t8 = torch.randn(3, 16)
+    assert _pad_k(t8) is t8
