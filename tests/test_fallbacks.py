@@ -137,10 +137,14 @@ def test_fused_optimizer_state_roundtrip_cpu():
         o1.zero_grad()
         m1(torch.randn(3, 6)).pow(2).mean().backward()
         o1.step()
-    sd = o1.state_dict()
+    # round-trip through serialization like real resume does —
+    # Optimizer.state_dict() returns references to live state tensors, so a
+    # direct load would alias the momentum buffers between o1 and o2
+    import copy
+    sd = copy.deepcopy(o1.state_dict())
 
     m2 = torch.nn.Linear(6, 4)
-    m2.load_state_dict(m1.state_dict())
+    m2.load_state_dict(copy.deepcopy(m1.state_dict()))
     o2 = FusedSGD(m2.parameters(), lr=0.1, momentum=0.9)
     o2.load_state_dict(sd)
     x = torch.randn(3, 6)
