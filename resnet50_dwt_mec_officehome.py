@@ -103,12 +103,16 @@ def build_loaders(args, rank, world):
         samplers['tgt'] = dist_data.DistributedSampler(tgt, world, rank)
     src_loader = DataLoader(src, batch_size=args.source_batch_size,
                             shuffle='src' not in samplers, sampler=samplers.get('src'),
-                            num_workers=args.num_workers, drop_last=True)
+                            num_workers=args.num_workers, drop_last=True,
+                            pin_memory=torch.cuda.is_available(),
+                            persistent_workers=args.num_workers > 0)
     # note: the reference uses source_batch_size for the target loader too
     #  — kept for parity
     tgt_loader = DataLoader(tgt, batch_size=args.source_batch_size,
                             shuffle='tgt' not in samplers, sampler=samplers.get('tgt'),
-                            num_workers=args.num_workers, drop_last=True)
+                            num_workers=args.num_workers, drop_last=True,
+                            pin_memory=torch.cuda.is_available(),
+                            persistent_workers=args.num_workers > 0)
     test_loader = DataLoader(tgt_test, batch_size=args.test_batch_size,
                              shuffle=True, num_workers=args.num_workers)
     return src_loader, tgt_loader, test_loader

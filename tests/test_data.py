@@ -99,3 +99,13 @@ def test_synthetic_datasets():
     assert not torch.equal(img, aug)
     img2, _, label2 = oh[3]
     assert torch.equal(img, img2) and label == label2  # deterministic per index
+
+
+def test_resize_numpy_path():
+    arr = np.random.rand(30, 40, 1).astype(np.float32)
+    out = Resize((16, 16))(arr)
+    t = ToTensor()(out)
+    assert t.shape == (1, 16, 16)
+    rgb = (np.random.rand(20, 20, 3) * 255).astype(np.uint8)
+    out = Resize(10)(rgb)
+    assert ToTensor()(out).shape == (3, 10, 10)

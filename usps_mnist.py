@@ -117,10 +117,14 @@ def main(argv=None):
         samplers['tgt'] = dist_data.DistributedSampler(tgt_train, world, rank)
     src_loader = DataLoader(src_train, batch_size=args.source_batch_size,
                             shuffle='src' not in samplers, sampler=samplers.get('src'),
-                            num_workers=args.num_workers, drop_last=True)
+                            num_workers=args.num_workers, drop_last=True,
+                            pin_memory=torch.cuda.is_available(),
+                            persistent_workers=args.num_workers > 0)
     tgt_loader = DataLoader(tgt_train, batch_size=args.source_batch_size,
                             shuffle='tgt' not in samplers, sampler=samplers.get('tgt'),
-                            num_workers=args.num_workers, drop_last=True)
+                            num_workers=args.num_workers, drop_last=True,
+                            pin_memory=torch.cuda.is_available(),
+                            persistent_workers=args.num_workers > 0)
     test_loader = DataLoader(tgt_test, batch_size=args.test_batch_size,
                              shuffle=True, num_workers=args.num_workers)
 
