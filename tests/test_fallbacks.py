@@ -165,3 +165,27 @@ def test_mfma_gemm_pad_helper():
     assert p[:, 13:].abs().sum() == 0
     t8 = torch.randn(3, 16)
     assert _pad_k(t8) is t8
+
+
+@pytest.mark.parametrize("mode", ["chol", "zca"])
+def test_degenerate_inputs_no_nan(mode):
+    """Constant channels (zero covariance) and large magnitudes stay finite —
+    the eps shrinkage guarantees a PD matrix-function input."""
+    cfg = dict(parts=1, num_groups=2, eps=1e-3, momentum=0.1, training=True,
+               mode=mode, relu=False)
+    x_const = torch.ones(6, 8, 4, 4) * 3.0
+    y = Fdwt.WhitenMulti.apply(x_const, None, None, None, None, cfg)
+    assert torch.isfinite(y).all()
+    assert y.abs().max() < 1e-4  # centered constant -> ~zero output
+
+    x_big = torch.randn(6, 8, 4, 4) * 1e3
+    y2 = Fdwt.WhitenMulti.apply(x_big.requires_grad_(True), None, None, None,
+                                None, cfg)
+    assert torch.isfinite(y2).all()
+    y2.sum().backward()
+    assert torch.isfinite(x_big.grad).all()
+
+    # BN with zero variance
+    bcfg = dict(parts=1, eps=1e-5, momentum=0.1, training=True, relu=False)
+    yb = Fdwt.BatchNormMulti.apply(x_const, None, None, None, None, bcfg)
+    assert torch.isfinite(yb).all()
