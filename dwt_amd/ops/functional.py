@@ -555,6 +555,7 @@ def mec_loss(x: torch.Tensor, y: torch.Tensor) -> torch.Tensor:
         from ..kernels import dispatch
         if dispatch.available():
             return dispatch.mec_loss(x, y)
+        dispatch.require_or_warn("mec_loss")
     return MecLossFn.apply(x, y)
 
 
@@ -563,6 +564,7 @@ def entropy_loss(x: torch.Tensor) -> torch.Tensor:
         from ..kernels import dispatch
         if dispatch.available():
             return dispatch.entropy_loss(x)
+        dispatch.require_or_warn("entropy_loss")
     return EntropyLossFn.apply(x)
 
 
@@ -628,4 +630,5 @@ def ce_loss(logits: torch.Tensor, target: torch.Tensor) -> torch.Tensor:
         from ..kernels import dispatch
         if dispatch.available():
             return dispatch.ce_loss(logits, target)
+        dispatch.require_or_warn("ce_loss")
     return F.nll_loss(F.log_softmax(logits.float(), dim=1), target)

@@ -127,8 +127,9 @@ def conv2d_dgrad(dy: torch.Tensor, weight: torch.Tensor, in_shape,
 
 
 class _MFMAConv2dFn(torch.autograd.Function):
-    """Training conv: our MFMA implicit-GEMM fwd + dgrad; wgrad via the
-    library (MFMA wgrad kernel is the round-2 item)."""
+    """Training conv on the hand-written MFMA kernels: implicit-GEMM fwd,
+    dgrad (skipped when the input needs no grad, e.g. the stem) and wgrad.
+    DWT_AMD_WGRAD=aten routes wgrad through the library as an escape hatch."""
 
     @staticmethod
     def forward(ctx, x, weight, bias, stride, padding):
@@ -139,14 +140,23 @@ class _MFMAConv2dFn(torch.autograd.Function):
 
     @staticmethod
     def backward(ctx, dout):
+        import os
         x, weight = ctx.saved_tensors
         stride, padding, has_bias = ctx.meta
         dout = dout.contiguous(memory_format=torch.channels_last)
-        dx = conv2d_dgrad(dout, weight, x.shape, stride=stride, padding=padding)
-        _, dw, db = torch.ops.aten.convolution_backward(
-            dout, x, weight, [weight.shape[0]] if has_bias else None,
-            [stride, stride], [padding, padding], [1, 1], False, [0, 0], 1,
-            [False, True, has_bias])
+        dx = None
+        if ctx.needs_input_grad[0]:
+            dx = conv2d_dgrad(dout, weight, x.shape, stride=stride,
+                              padding=padding)
+        if os.environ.get("DWT_AMD_WGRAD") == "aten":
+            _, dw, db = torch.ops.aten.convolution_backward(
+                dout, x, weight, [weight.shape[0]] if has_bias else None,
+                [stride, stride], [padding, padding], [1, 1], False, [0, 0], 1,
+                [False, True, has_bias])
+        else:
+            dw = conv2d_wgrad(dout, x, tuple(weight.shape), stride=stride,
+                              padding=padding)
+            db = dout.sum(dim=(0, 2, 3)).to(dout.dtype) if has_bias else None
         return dx, dw, db, None, None
 
 

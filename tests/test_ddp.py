@@ -106,6 +106,7 @@ def test_bucket_assembly():
     ddp.params = [p for p in model.parameters() if p.requires_grad]
     ddp.buckets = []
     ddp._param_bucket = {}
+    ddp.reduce_dtype = None
     ddp._build_buckets(0.1)
     seen = set()
     for b in ddp.buckets:
@@ -131,6 +132,84 @@ def test_digits_entrypoint_torchrun_cpu(tmp_path):
         cwd=repo, capture_output=True, text=True, timeout=900)
     assert r.returncode == 0, (r.stdout[-800:], r.stderr[-1500:])
     assert "Test set" in r.stdout
+
+
+def _run_grad_parity_dtype(rank, world, port, results, dtype_name, reduce_fp32):
+    """Grad parity on a small MLP at configurable world size / dtype /
+    reduce mode, including the steady-state grad-as-bucket-view step
+    (second backward accumulates directly into the bucket buffer)."""
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    try:
+        from dwt_amd.parallel import BucketedDataParallel
+        dtype = dict(float32=torch.float32, bfloat16=torch.bfloat16)[dtype_name]
+        torch.manual_seed(0)
+        model = torch.nn.Sequential(
+            torch.nn.Linear(32, 64), torch.nn.ReLU(),
+            torch.nn.Linear(64, 64), torch.nn.ReLU(),
+            torch.nn.Linear(64, 8)).to(dtype)
+        clone = torch.nn.Sequential(
+            torch.nn.Linear(32, 64), torch.nn.ReLU(),
+            torch.nn.Linear(64, 64), torch.nn.ReLU(),
+            torch.nn.Linear(64, 8)).to(dtype)
+        clone.load_state_dict(model.state_dict())
+        ddp = BucketedDataParallel(
+            model, bucket_cap_mb=0.01,
+            reduce_dtype=torch.float32 if reduce_fp32 else None)
+
+        for it in range(2):  # step 2 exercises the aliased-grad fast path
+            torch.manual_seed(1000 * it + rank)
+            x = torch.randn(8, 32).to(dtype)
+            model(x).float().pow(2).mean().backward()
+            clone(x).float().pow(2).mean().backward()
+            ddp.sync()
+            for (n, p), pc in zip(model.named_parameters(), clone.parameters()):
+                avg = pc.grad.detach().float().clone()
+                dist.all_reduce(avg)
+                avg /= world
+                tol = 1e-6 if dtype == torch.float32 else 3e-2
+                assert torch.allclose(p.grad.float(), avg, atol=tol,
+                                      rtol=tol), (it, n)
+            if not reduce_fp32:
+                # steady state: grads must alias the bucket buffers
+                for b in ddp.buckets:
+                    for p in b.params:
+                        assert p.grad is b.views[p]
+            # zero in place (FusedSGD-style) so the aliasing survives into
+            # the next iteration on the fast path
+            for p in model.parameters():
+                if reduce_fp32:
+                    p.grad = None
+                elif p.grad is not None:
+                    p.grad.detach().zero_()
+            for p in clone.parameters():
+                p.grad = None
+        if rank == 0:
+            results.put("ok")
+    finally:
+        dist.destroy_process_group()
+
+
+@pytest.mark.parametrize("world,dtype_name,reduce_fp32", [
+    (2, "float32", False),
+    (4, "float32", False),
+    (4, "bfloat16", False),   # pre-divided bf16 SUM (ADVICE r1 item 2)
+    (2, "bfloat16", True),    # fp32 bucket mode
+])
+def test_ddp_grad_parity_modes(world, dtype_name, reduce_fp32):
+    port = 29750 + abs(hash((world, dtype_name, reduce_fp32))) % 200
+    ctx = mp.get_context("spawn")
+    results = ctx.Queue()
+    procs = [ctx.Process(target=_run_grad_parity_dtype,
+                         args=(r, world, port, results, dtype_name, reduce_fp32))
+             for r in range(world)]
+    for p in procs:
+        p.start()
+    for p in procs:
+        p.join(timeout=240)
+    assert all(p.exitcode == 0 for p in procs), [p.exitcode for p in procs]
+    assert results.get(timeout=5) == "ok"
 
 
 def _run_stats_sync_parity(rank, world, port, results):

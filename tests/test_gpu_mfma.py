@@ -141,10 +141,6 @@ def test_mfma_conv2d_module_trains(dev):
     assert _rel_err(conv.weight.grad, ref.weight.grad) < 0.05
 
 
-@pytest.mark.skip(reason="wgrad kernel landed after round-1 GPU budget was "
-                         "exhausted; index math is CPU-validated "
-                         "(tests/test_wgrad_indexing.py) — unskip and "
-                         "validate on-GPU first thing in round 2")
 @pytest.mark.parametrize("shape", [
     (4, 64, 28, 28, 64, 1, 1, 0),
     (4, 64, 28, 28, 64, 3, 1, 1),
