@@ -60,6 +60,18 @@ def main():
              "| shape | M x N x K | GFLOP | ours ms | lib ms | ours TF | lib TF | ratio |",
              "|---|---|---|---|---|---|---|---|"]
 
+    from dwt_amd.ops.mfma import conv2d_dgrad, conv2d_wgrad
+
+    def lib_dgrad(g, x, wt, stride, pad):
+        return torch.ops.aten.convolution_backward(
+            g, x, wt, None, [stride, stride], [pad, pad], [1, 1], False,
+            [0, 0], 1, [True, False, False])[0]
+
+    def lib_wgrad(g, x, wt, stride, pad):
+        return torch.ops.aten.convolution_backward(
+            g, x, wt, None, [stride, stride], [pad, pad], [1, 1], False,
+            [0, 0], 1, [False, True, False])[1]
+
     for name, n, cin, h, w, cout, k, stride, pad in R50_SHAPES:
         x = torch.randn(n, cin, h, w, device=dev).to(torch.bfloat16) \
             .contiguous(memory_format=torch.channels_last)
@@ -68,14 +80,27 @@ def main():
         p = (h + 2 * pad - k) // stride + 1
         m = n * p * p
         kk = k * k * cin
+        g = torch.randn(n, cout, p, p, device=dev).to(torch.bfloat16) \
+            .contiguous(memory_format=torch.channels_last)
         gflop = 2.0 * m * cout * kk / 1e9
-        t_ours = timeit(lambda: conv2d_fwd(x, wt, stride=stride, padding=pad), args.iters)
-        t_lib = timeit(lambda: F.conv2d(x, wt, stride=stride, padding=pad), args.iters)
-        lines.append(f"| {name} | {m}x{cout}x{kk} | {gflop:.1f} | "
-                     f"{t_ours*1e3:.3f} | {t_lib*1e3:.3f} | "
-                     f"{gflop/t_ours/1e3:.0f} | {gflop/t_lib/1e3:.0f} | "
-                     f"{t_lib/t_ours:.2f}x |")
-        print(lines[-1], flush=True)
+        passes = [
+            ("fwd", lambda: conv2d_fwd(x, wt, stride=stride, padding=pad),
+             lambda: F.conv2d(x, wt, stride=stride, padding=pad)),
+            ("dgrad", lambda: conv2d_dgrad(g, wt, x.shape, stride=stride, padding=pad),
+             lambda: lib_dgrad(g, x, wt, stride, pad)),
+            ("wgrad", lambda: conv2d_wgrad(g, x, tuple(wt.shape), stride=stride, padding=pad),
+             lambda: lib_wgrad(g, x, wt, stride, pad)),
+        ]
+        for pname, ours_fn, lib_fn in passes:
+            if pname == "dgrad" and name == "stem7x7":
+                continue  # stem never needs dx
+            t_ours = timeit(ours_fn, args.iters)
+            t_lib = timeit(lib_fn, args.iters)
+            lines.append(f"| {name}:{pname} | {m}x{cout}x{kk} | {gflop:.1f} | "
+                         f"{t_ours*1e3:.3f} | {t_lib*1e3:.3f} | "
+                         f"{gflop/t_ours/1e3:.0f} | {gflop/t_lib/1e3:.0f} | "
+                         f"{t_lib/t_ours:.2f}x |")
+            print(lines[-1], flush=True)
 
     for name, m, k, n in GEMM_SHAPES:
         a = torch.randn(m, k, device=dev).to(torch.bfloat16)
