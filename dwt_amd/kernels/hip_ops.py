@@ -65,6 +65,13 @@ def _f32c(t):
 
 
 class _HipWhitenMulti(torch.autograd.Function):
+    """All per-branch statistics are STACKED ([parts*C] means, [parts*G, g, g]
+    covariances/W) so the channels_last path runs ONE launch per pass across
+    all domain branches (grid.y = branch — late-layer sites are launch-bound
+    otherwise, profiles/norm_bench.md); the matrix-function kernels always
+    see the stacked group axis.  The NCHW path loops branches over contiguous
+    slices of the same stacked tensors."""
+
     @staticmethod
     def forward(ctx, x, gamma, beta, running_means, running_vars, cfg):
         ext = _ext()
@@ -84,6 +91,7 @@ class _HipWhitenMulti(torch.autograd.Function):
         assert n % parts == 0,             f"batch {n} not divisible by {parts} domain branches"
         b = n // parts
         n_groups = c // g
+        ng_all = parts * n_groups
         dev = x.device
         m_count = b * h * w
 
@@ -92,61 +100,67 @@ class _HipWhitenMulti(torch.autograd.Function):
         bflat = beta.detach().reshape(c).contiguous() if has_affine else torch.empty(0, device=dev, dtype=x.dtype)
 
         out = torch.empty_like(x)
-        means, wmats, saved_mat = [], [], []
-        for p in range(parts):
-            xp = x[p * b:(p + 1) * b]
-            if use_batch:
-                acc = torch.zeros(n_groups * (g + g * g), device=dev, dtype=torch.float32)
-                mean = torch.empty(c, device=dev, dtype=torch.float32)
-                cov = torch.empty(n_groups, g, g, device=dev, dtype=torch.float32)
-                if layout == "cl":
-                    ext.whiten_stats_cl(xp, acc, mean, cov, g, c, b * h * w)
-                else:
-                    ext.whiten_stats(xp, acc, mean, cov, g)
-            else:
-                mean = _f32c(running_means[p]).reshape(c)
-                cov = _f32c(running_vars[p]).reshape(n_groups, g, g)
-
-            wmat = torch.empty(n_groups, g, g, device=dev, dtype=torch.float32)
-            if mode == "chol":
-                ell = torch.empty_like(wmat)
-                ext.matfn_chol_fwd(cov, wmat, ell, eps)
-                saved_mat.append(ell)
-            else:
-                ys = torch.empty(n_groups, ns_iters, g, g, device=dev, dtype=torch.float32)
-                zs = torch.empty_like(ys)
-                svals = torch.empty(n_groups, device=dev, dtype=torch.float32)
-                ext.matfn_ns_fwd(cov, wmat, ys, zs, svals, eps, ns_iters)
-                saved_mat.append((ys, zs, svals))
-
+        if use_batch:
+            acc = torch.zeros(ng_all * (g + g * g), device=dev, dtype=torch.float32)
+            mean = torch.empty(parts * c, device=dev, dtype=torch.float32)
+            cov = torch.empty(ng_all, g, g, device=dev, dtype=torch.float32)
             if layout == "cl":
-                ext.whiten_apply_cl(xp, mean, wmat, gflat, bflat,
-                                    out[p * b:(p + 1) * b], g, c, b * h * w,
-                                    relu, has_affine)
+                ext.whiten_stats_cl(x, acc, mean, cov, g, c, m_count, parts)
             else:
-                ext.whiten_apply(xp, mean, wmat, gflat, bflat,
-                                 out[p * b:(p + 1) * b], g, relu, has_affine)
-            means.append(mean)
-            wmats.append(wmat)
+                for p in range(parts):
+                    ext.whiten_stats(x[p * b:(p + 1) * b],
+                                     acc[p * n_groups * (g + g * g):(p + 1) * n_groups * (g + g * g)],
+                                     mean[p * c:(p + 1) * c],
+                                     cov[p * n_groups:(p + 1) * n_groups], g)
+        else:
+            mean = torch.stack([_f32c(running_means[p]).reshape(c)
+                                for p in range(parts)]).reshape(-1).contiguous()
+            cov = torch.stack([_f32c(running_vars[p]).reshape(n_groups, g, g)
+                               for p in range(parts)]).reshape(ng_all, g, g).contiguous()
 
-            if training and track and running_means is not None:
-                with torch.no_grad():
+        wmat = torch.empty(ng_all, g, g, device=dev, dtype=torch.float32)
+        if mode == "chol":
+            ell = torch.empty_like(wmat)
+            ext.matfn_chol_fwd(cov, wmat, ell, eps)
+            saved_mat = ell
+        else:
+            ys = torch.empty(ng_all, ns_iters, g, g, device=dev, dtype=torch.float32)
+            zs = torch.empty_like(ys)
+            svals = torch.empty(ng_all, device=dev, dtype=torch.float32)
+            ext.matfn_ns_fwd(cov, wmat, ys, zs, svals, eps, ns_iters)
+            saved_mat = (ys, zs, svals)
+
+        if layout == "cl":
+            ext.whiten_apply_cl(x, mean, wmat, gflat, bflat, out, g, c,
+                                m_count, relu, has_affine, parts)
+        else:
+            for p in range(parts):
+                ext.whiten_apply(x[p * b:(p + 1) * b], mean[p * c:(p + 1) * c],
+                                 wmat[p * n_groups:(p + 1) * n_groups], gflat,
+                                 bflat, out[p * b:(p + 1) * b], g, relu,
+                                 has_affine)
+
+        if training and track and running_means is not None:
+            with torch.no_grad():
+                for p in range(parts):
                     rm, rv = running_means[p], running_vars[p]
+                    mslice = mean[p * c:(p + 1) * c]
+                    cslice = cov[p * n_groups:(p + 1) * n_groups]
                     if rm.dtype == torch.float32 and rm.is_contiguous() \
                             and rv.dtype == torch.float32 and rv.is_contiguous():
-                        ext.ema_update(rm, mean, rv, cov, momentum)
+                        ext.ema_update(rm, mslice, rv, cslice, momentum)
                     else:
                         rm.mul_(1.0 - momentum).add_(
-                            mean.reshape(rm.shape).to(rm.dtype), alpha=momentum)
+                            mslice.reshape(rm.shape).to(rm.dtype), alpha=momentum)
                         rv.mul_(1.0 - momentum).add_(
-                            cov.reshape(rv.shape).to(rv.dtype), alpha=momentum)
+                            cslice.reshape(rv.shape).to(rv.dtype), alpha=momentum)
 
         ctx.cfg = cfg
         ctx.g = g
         ctx.layout = layout
         ctx.m_count = m_count
-        ctx.means = means
-        ctx.wmats = wmats
+        ctx.mean = mean
+        ctx.wmat = wmat
         ctx.saved_mat = saved_mat
         ctx.has_affine = has_affine
         ctx.gflat = gflat
@@ -167,51 +181,65 @@ class _HipWhitenMulti(torch.autograd.Function):
         n, c, h, w = x.shape
         b = n // parts
         n_groups = c // g
+        ng_all = parts * n_groups
         dev = x.device
         layout = ctx.layout
         if layout == "cl":
             dout = dout.contiguous(memory_format=torch.channels_last)
         else:
             dout = dout.contiguous()
+        mean, wmat = ctx.mean, ctx.wmat
+        m_count = ctx.m_count
 
         dgb = torch.zeros(parts, 2, c, device=dev, dtype=torch.float32)
+        dW = torch.zeros(ng_all, g, g, device=dev, dtype=torch.float32)
         dx = torch.empty_like(x)
-        for p in range(parts):
-            sl = slice(p * b, (p + 1) * b)
-            xp, doutp, outp = x[sl], dout[sl], out[sl]
-            mean, wmat = ctx.means[p], ctx.wmats[p]
-            dW = torch.zeros(n_groups, g, g, device=dev, dtype=torch.float32)
-            if layout == "cl":
-                ext.whiten_bwd_reduce_cl(xp, doutp, outp, mean, wmat, ctx.gflat,
-                                         dW, dgb[p].reshape(-1), g, c, b * h * w,
-                                         relu, ctx.has_affine)
+
+        if layout == "cl":
+            ext.whiten_bwd_reduce_cl(x, dout, out, mean, wmat, ctx.gflat,
+                                     dW, dgb.reshape(-1), g, c, m_count,
+                                     relu, ctx.has_affine, parts)
+        else:
+            for p in range(parts):
+                sl = slice(p * b, (p + 1) * b)
+                ext.whiten_bwd_reduce(x[sl], dout[sl], out[sl],
+                                      mean[p * c:(p + 1) * c],
+                                      wmat[p * n_groups:(p + 1) * n_groups],
+                                      ctx.gflat, dW[p * n_groups:(p + 1) * n_groups],
+                                      dgb[p].reshape(-1), g, relu, ctx.has_affine)
+
+        if use_batch:
+            gdb = (ctx.gflat.float().unsqueeze(0) * dgb[:, 1]) if ctx.has_affine \
+                else dgb[:, 1]
+            gdb = gdb.reshape(-1).contiguous()
+            S = torch.empty(ng_all, g, g, device=dev, dtype=torch.float32)
+            corr = torch.empty(parts * c, device=dev, dtype=torch.float32)
+            inv_m = 1.0 / m_count
+            if mode == "chol":
+                ext.matfn_chol_bwd(dW, wmat, ctx.saved_mat, gdb, S, corr,
+                                   eps, inv_m)
             else:
-                ext.whiten_bwd_reduce(xp, doutp, outp, mean, wmat, ctx.gflat,
-                                      dW, dgb[p].reshape(-1), g, relu, ctx.has_affine)
-            if use_batch:
-                gdb = (ctx.gflat.float() * dgb[p, 1]) if ctx.has_affine else dgb[p, 1]
-                gdb = gdb.contiguous()
-                S = torch.empty(n_groups, g, g, device=dev, dtype=torch.float32)
-                corr = torch.empty(c, device=dev, dtype=torch.float32)
-                inv_m = 1.0 / ctx.m_count
-                if mode == "chol":
-                    ext.matfn_chol_bwd(dW, wmat, ctx.saved_mat[p], gdb, S, corr,
-                                       eps, inv_m)
-                else:
-                    ys, zs, svals = ctx.saved_mat[p]
-                    ext.matfn_ns_bwd(dW, wmat, ys, zs, svals, gdb, S, corr,
-                                     eps, inv_m, ns_iters)
-            else:
-                S = torch.empty(0, device=dev)
-                corr = torch.empty(0, device=dev)
-            if layout == "cl":
-                ext.whiten_bwd_apply_cl(xp, doutp, outp, mean, wmat, ctx.gflat,
-                                        S, corr, dx[sl], g, c, b * h * w, relu,
-                                        ctx.has_affine, use_batch)
-            else:
-                ext.whiten_bwd_apply(xp, doutp, outp, mean, wmat, ctx.gflat,
-                                     S, corr, dx[sl], g, relu, ctx.has_affine,
-                                     use_batch)
+                ys, zs, svals = ctx.saved_mat
+                ext.matfn_ns_bwd(dW, wmat, ys, zs, svals, gdb, S, corr,
+                                 eps, inv_m, ns_iters)
+        else:
+            S = torch.empty(0, device=dev)
+            corr = torch.empty(0, device=dev)
+
+        if layout == "cl":
+            ext.whiten_bwd_apply_cl(x, dout, out, mean, wmat, ctx.gflat,
+                                    S, corr, dx, g, c, m_count, relu,
+                                    ctx.has_affine, use_batch, parts)
+        else:
+            for p in range(parts):
+                sl = slice(p * b, (p + 1) * b)
+                Sp = S[p * n_groups:(p + 1) * n_groups] if use_batch else S
+                cp = corr[p * c:(p + 1) * c] if use_batch else corr
+                ext.whiten_bwd_apply(x[sl], dout[sl], out[sl],
+                                     mean[p * c:(p + 1) * c],
+                                     wmat[p * n_groups:(p + 1) * n_groups],
+                                     ctx.gflat, Sp, cp, dx[sl], g, relu,
+                                     ctx.has_affine, use_batch)
 
         if ctx.has_affine:
             dgamma = dgb[:, 0].sum(0).reshape(gamma.shape).to(gamma.dtype)
@@ -258,46 +286,54 @@ class _HipBatchNormMulti(torch.autograd.Function):
         bflat = beta.detach().reshape(c).contiguous() if has_affine else torch.empty(0, device=dev, dtype=x.dtype)
 
         out = torch.empty_like(x)
-        means, istds = [], []
-        for p in range(parts):
-            xp = x[p * b:(p + 1) * b]
-            if use_batch:
-                acc = torch.zeros(2 * c, device=dev, dtype=torch.float32)
-                mean = torch.empty(c, device=dev, dtype=torch.float32)
-                istd = torch.empty(c, device=dev, dtype=torch.float32)
-                var_unb = torch.empty(c, device=dev, dtype=torch.float32)
-                if layout == "cl":
-                    ext.bn_stats_cl(xp, acc, mean, istd, var_unb, c, cnt, eps)
-                else:
-                    ext.bn_stats(xp, acc, mean, istd, var_unb, eps)
-                if training and track and running_means is not None:
-                    with torch.no_grad():
+        if use_batch:
+            acc = torch.zeros(parts * 2 * c, device=dev, dtype=torch.float32)
+            mean = torch.empty(parts * c, device=dev, dtype=torch.float32)
+            istd = torch.empty(parts * c, device=dev, dtype=torch.float32)
+            var_unb = torch.empty(parts * c, device=dev, dtype=torch.float32)
+            if layout == "cl":
+                ext.bn_stats_cl(x, acc, mean, istd, var_unb, c, cnt, eps, parts)
+            else:
+                for p in range(parts):
+                    ext.bn_stats(x[p * b:(p + 1) * b],
+                                 acc[p * 2 * c:(p + 1) * 2 * c],
+                                 mean[p * c:(p + 1) * c],
+                                 istd[p * c:(p + 1) * c],
+                                 var_unb[p * c:(p + 1) * c], eps)
+            if training and track and running_means is not None:
+                with torch.no_grad():
+                    for p in range(parts):
                         rm, rv = running_means[p], running_vars[p]
+                        mslice = mean[p * c:(p + 1) * c]
+                        vslice = var_unb[p * c:(p + 1) * c]
                         if rm.dtype == torch.float32 and rm.is_contiguous() \
                                 and rv.dtype == torch.float32 and rv.is_contiguous():
-                            ext.ema_update(rm, mean, rv, var_unb, momentum)
+                            ext.ema_update(rm, mslice, rv, vslice, momentum)
                         else:
-                            rm.mul_(1 - momentum).add_(mean.to(rm.dtype), alpha=momentum)
-                            rv.mul_(1 - momentum).add_(var_unb.to(rv.dtype), alpha=momentum)
-            else:
-                mean = _f32c(running_means[p]).reshape(c)
-                var = _f32c(running_vars[p]).reshape(c)
-                istd = torch.rsqrt(var + eps)
-            if layout == "cl":
-                ext.bn_apply_cl(xp, mean, istd, gflat, bflat,
-                                out[p * b:(p + 1) * b], c, cnt, relu, has_affine)
-            else:
-                ext.bn_apply(xp, mean, istd, gflat, bflat,
+                            rm.mul_(1 - momentum).add_(mslice.to(rm.dtype), alpha=momentum)
+                            rv.mul_(1 - momentum).add_(vslice.to(rv.dtype), alpha=momentum)
+        else:
+            mean = torch.stack([_f32c(running_means[p]).reshape(c)
+                                for p in range(parts)]).reshape(-1).contiguous()
+            var = torch.stack([_f32c(running_vars[p]).reshape(c)
+                               for p in range(parts)]).reshape(-1)
+            istd = torch.rsqrt(var + eps).contiguous()
+
+        if layout == "cl":
+            ext.bn_apply_cl(x, mean, istd, gflat, bflat, out, c, cnt, relu,
+                            has_affine, parts)
+        else:
+            for p in range(parts):
+                ext.bn_apply(x[p * b:(p + 1) * b], mean[p * c:(p + 1) * c],
+                             istd[p * c:(p + 1) * c], gflat, bflat,
                              out[p * b:(p + 1) * b], relu, has_affine)
-            means.append(mean)
-            istds.append(istd)
 
         ctx.cfg = cfg
         ctx.layout = layout
         ctx.cnt = cnt
         ctx.spatial = spatial
-        ctx.means = means
-        ctx.istds = istds
+        ctx.mean = mean
+        ctx.istd = istd
         ctx.has_affine = has_affine
         ctx.gflat = gflat
         ctx.use_batch = use_batch
@@ -318,22 +354,26 @@ class _HipBatchNormMulti(torch.autograd.Function):
             dout = dout.contiguous(memory_format=torch.channels_last)
         else:
             dout = dout.contiguous()
+        mean, istd = ctx.mean, ctx.istd
 
         sums = torch.zeros(parts, 2, c, device=dev, dtype=torch.float32)
         dx = torch.empty_like(x)
-        for p in range(parts):
-            sl = slice(p * b, (p + 1) * b)
-            xp, doutp, outp = x[sl], dout[sl], out[sl]
-            if layout == "cl":
-                ext.bn_bwd_reduce_cl(xp, doutp, outp, ctx.means[p], ctx.istds[p],
-                                     sums[p].reshape(-1), c, ctx.cnt, relu)
-                ext.bn_bwd_apply_cl(xp, doutp, outp, ctx.means[p], ctx.istds[p],
-                                    ctx.gflat, sums[p].reshape(-1), dx[sl], c,
-                                    ctx.cnt, relu, ctx.has_affine, ctx.use_batch)
-            else:
-                ext.bn_bwd_reduce(xp, doutp, outp, ctx.means[p], ctx.istds[p],
+        if layout == "cl":
+            ext.bn_bwd_reduce_cl(x, dout, out, mean, istd, sums.reshape(-1),
+                                 c, ctx.cnt, relu, parts)
+            ext.bn_bwd_apply_cl(x, dout, out, mean, istd, ctx.gflat,
+                                sums.reshape(-1), dx, c, ctx.cnt, relu,
+                                ctx.has_affine, ctx.use_batch, parts)
+        else:
+            for p in range(parts):
+                sl = slice(p * b, (p + 1) * b)
+                ext.bn_bwd_reduce(x[sl], dout[sl], out[sl],
+                                  mean[p * c:(p + 1) * c],
+                                  istd[p * c:(p + 1) * c],
                                   sums[p].reshape(-1), relu)
-                ext.bn_bwd_apply(xp, doutp, outp, ctx.means[p], ctx.istds[p],
+                ext.bn_bwd_apply(x[sl], dout[sl], out[sl],
+                                 mean[p * c:(p + 1) * c],
+                                 istd[p * c:(p + 1) * c],
                                  ctx.gflat, sums[p].reshape(-1), dx[sl], relu,
                                  ctx.has_affine, ctx.use_batch)
 

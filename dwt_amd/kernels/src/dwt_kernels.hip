@@ -805,15 +805,19 @@ template <> DEV_INLINE void store_group<c10::BFloat16, 8>(c10::BFloat16* p, cons
                  f32x2_to_bf16(i[4], i[5]), f32x2_to_bf16(i[6], i[7]));
 }
 
-// fused mean+cov, NHWC
+// fused mean+cov, NHWC.  blockIdx.y = domain branch ("part"): x advances by
+// part*M*C and the per-part statistics are stacked ([parts, ...]) — one
+// launch covers all three domain branches of a norm site (launch-bound at
+// 7x7/14x14 otherwise; see profiles/norm_bench.md).
 template <typename T, int G>
 __global__ void whiten_stats_nhwc_kernel(
     const T* __restrict__ x, float* __restrict__ acc,
-    int C, int64_t M /* = N*H*W positions */) {
+    int C, int64_t M /* = per-part N*H*W positions */) {
+  const int part = blockIdx.y;
+  x += (int64_t)part * M * C;
   const int CW = C < 256 ? C : 256;
   const int GW = CW / G;
   const int c0 = blockIdx.z * 256 + (threadIdx.x % GW) * G;
-  const int grp = c0 / G;
   const int rows_per_iter = blockDim.x / GW;
   const int row_in_block = threadIdx.x / GW;
 
@@ -860,7 +864,7 @@ __global__ void whiten_stats_nhwc_kernel(
   for (int k = threadIdx.x; k < GW * (G + NTRI); k += blockDim.x) {
     const int lg = k / (G + NTRI);
     const int kk = k % (G + NTRI);
-    const int gidx = blockIdx.z * (256 / G) + lg;
+    const int gidx = part * (C / G) + blockIdx.z * (256 / G) + lg;
     float* gacc = acc + (int64_t)gidx * (G + G * G);
     if (kk < G) {
       atomicAdd(&gacc[kk], lacc[k]);
@@ -880,10 +884,14 @@ __global__ void whiten_apply_nhwc_kernel(
     const float* __restrict__ W, const T* __restrict__ gamma,
     const T* __restrict__ beta, T* __restrict__ out,
     int C, int64_t M, int relu, int has_affine) {
+  const int part = blockIdx.y;
+  x += (int64_t)part * M * C;
+  out += (int64_t)part * M * C;
+  mean += (int64_t)part * C;
   const int CW = C < 256 ? C : 256;
   const int GW = CW / G;
   const int c0 = blockIdx.z * 256 + (threadIdx.x % GW) * G;
-  const int grp = c0 / G;
+  const int grp = part * (C / G) + c0 / G;
   const int rows_per_iter = blockDim.x / GW;
   const int row_in_block = threadIdx.x / GW;
 
@@ -923,10 +931,15 @@ __global__ void whiten_bwd_reduce_nhwc_kernel(
     const float* __restrict__ W, const T* __restrict__ gamma,
     float* __restrict__ dWacc, float* __restrict__ dgb,
     int C, int64_t M, int relu, int has_affine) {
+  const int part = blockIdx.y;
+  const int64_t poff = (int64_t)part * M * C;
+  x += poff; dout += poff; out += poff;
+  mean += (int64_t)part * C;
+  dgb += (int64_t)part * 2 * C;
   const int CW = C < 256 ? C : 256;
   const int GW = CW / G;
   const int c0 = blockIdx.z * 256 + (threadIdx.x % GW) * G;
-  const int grp = c0 / G;
+  const int grp = part * (C / G) + c0 / G;
   const int rows_per_iter = blockDim.x / GW;
   const int row_in_block = threadIdx.x / GW;
 
@@ -988,8 +1001,9 @@ __global__ void whiten_bwd_reduce_nhwc_kernel(
   for (int k = threadIdx.x; k < GW * NV; k += blockDim.x) {
     const int lg = k / NV;
     const int kk = k % NV;
-    const int gidx = blockIdx.z * (256 / G) + lg;
-    const int cg0 = gidx * G;
+    const int lgidx = blockIdx.z * (256 / G) + lg;      // group within part
+    const int gidx = part * (C / G) + lgidx;            // global group
+    const int cg0 = lgidx * G;
     if (kk < G) atomicAdd(&dgb[cg0 + kk], lacc[k]);
     else if (kk < 2 * G) atomicAdd(&dgb[C + cg0 + (kk - G)], lacc[k]);
     else atomicAdd(&dWacc[(int64_t)gidx * G * G + (kk - 2 * G)], lacc[k]);
@@ -1005,10 +1019,15 @@ __global__ void whiten_bwd_apply_nhwc_kernel(
     const float* __restrict__ S, const float* __restrict__ corr,
     T* __restrict__ dx, int C, int64_t M, int relu, int has_affine,
     int train_stats) {
+  const int part = blockIdx.y;
+  const int64_t poff = (int64_t)part * M * C;
+  x += poff; dout += poff; out += poff; dx += poff;
+  mean += (int64_t)part * C;
+  corr += (int64_t)part * C;
   const int CW = C < 256 ? C : 256;
   const int GW = CW / G;
   const int c0 = blockIdx.z * 256 + (threadIdx.x % GW) * G;
-  const int grp = c0 / G;
+  const int grp = part * (C / G) + c0 / G;
   const int rows_per_iter = blockDim.x / GW;
   const int row_in_block = threadIdx.x / GW;
 
@@ -1051,10 +1070,14 @@ __global__ void whiten_bwd_apply_nhwc_kernel(
   }
 }
 
-// BN NHWC: per-lane 4-channel chunks
+// BN NHWC: per-lane 4-channel chunks.  blockIdx.y = domain branch, as in the
+// whitening NHWC family (acc/sums/mean/istd stacked [parts, ...]).
 template <typename T>
 __global__ void bn_stats_nhwc_kernel(
     const T* __restrict__ x, float* __restrict__ acc, int C, int64_t M) {
+  const int part = blockIdx.y;
+  x += (int64_t)part * M * C;
+  acc += (int64_t)part * 2 * C;
   constexpr int VC = 4;
   const int CW = C < 1024 ? C : 1024;
   const int NCH = CW / VC;
@@ -1095,6 +1118,11 @@ __global__ void bn_apply_nhwc_kernel(
     const float* __restrict__ istd, const T* __restrict__ gamma,
     const T* __restrict__ beta, T* __restrict__ out, int C, int64_t M,
     int relu, int has_affine) {
+  const int part = blockIdx.y;
+  x += (int64_t)part * M * C;
+  out += (int64_t)part * M * C;
+  mean += (int64_t)part * C;
+  istd += (int64_t)part * C;
   constexpr int VC = 4;
   const int CW = C < 1024 ? C : 1024;
   const int NCH = CW / VC;
@@ -1129,6 +1157,12 @@ __global__ void bn_bwd_reduce_nhwc_kernel(
     const T* __restrict__ out, const float* __restrict__ mean,
     const float* __restrict__ istd, float* __restrict__ sums, int C,
     int64_t M, int relu) {
+  const int part = blockIdx.y;
+  const int64_t poff = (int64_t)part * M * C;
+  x += poff; dout += poff; out += poff;
+  mean += (int64_t)part * C;
+  istd += (int64_t)part * C;
+  sums += (int64_t)part * 2 * C;
   constexpr int VC = 4;
   const int CW = C < 1024 ? C : 1024;
   const int NCH = CW / VC;
@@ -1178,6 +1212,12 @@ __global__ void bn_bwd_apply_nhwc_kernel(
     const float* __restrict__ istd, const T* __restrict__ gamma,
     const float* __restrict__ sums, T* __restrict__ dx, int C, int64_t M,
     float inv_m, int relu, int has_affine, int use_batch) {
+  const int part = blockIdx.y;
+  const int64_t poff = (int64_t)part * M * C;
+  x += poff; dout += poff; out += poff; dx += poff;
+  mean += (int64_t)part * C;
+  istd += (int64_t)part * C;
+  sums += (int64_t)part * 2 * C;
   constexpr int VC = 4;
   const int CW = C < 1024 ? C : 1024;
   const int NCH = CW / VC;
@@ -1253,17 +1293,20 @@ __global__ void bn_stats_partial_kernel(
   });
 }
 
+// total = parts*C channels; acc layout [parts][2][C], outputs flat [parts*C]
 __global__ void bn_stats_final_kernel(
     const float* __restrict__ acc, float* __restrict__ mean,
     float* __restrict__ istd, float* __restrict__ var_unb,
-    int C, float inv_m, float unb_scale, float eps) {
-  const int c = blockIdx.x * blockDim.x + threadIdx.x;
-  if (c >= C) return;
-  const float mu = acc[c] * inv_m;
-  float var = fmaxf(acc[C + c] * inv_m - mu * mu, 0.f);
-  mean[c] = mu;
-  istd[c] = rsqrtf(var + eps);
-  var_unb[c] = var * unb_scale;
+    int total, int Cper, float inv_m, float unb_scale, float eps) {
+  const int i = blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= total) return;
+  const int p = i / Cper, c = i % Cper;
+  const int64_t base = (int64_t)p * 2 * Cper;
+  const float mu = acc[base + c] * inv_m;
+  float var = fmaxf(acc[base + Cper + c] * inv_m - mu * mu, 0.f);
+  mean[i] = mu;
+  istd[i] = rsqrtf(var + eps);
+  var_unb[i] = var * unb_scale;
 }
 
 template <typename T, bool VECTOR>
@@ -1468,6 +1511,84 @@ __global__ void maxpool_nhwc_bwd_kernel(
     }
   }
   stf(dx + i, acc);
+}
+
+// vectorized variants: one thread owns a VC-channel chunk (16/32-B group
+// loads); 8x fewer threads and full-width accesses vs the scalar kernels
+// (measured 10.5 ms/step scalar at B=1536 — pure gather/scatter bound).
+template <typename T, int VC>
+__global__ void maxpool_nhwc_fwd_vec_kernel(
+    const T* __restrict__ x, T* __restrict__ out, uint8_t* __restrict__ idx,
+    int C, int H, int W, int P, int Q, int KS, int stride, int pad,
+    int64_t total_chunks) {  // total = N*P*Q*(C/VC)
+  const int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= total_chunks) return;
+  const int nc = C / VC;
+  const int c0 = (int)(i % nc) * VC;
+  int64_t r = i / nc;
+  const int q = r % Q; r /= Q;
+  const int p = r % P; r /= P;
+  const int n = (int)r;
+  float best[VC];
+  uint8_t bidx[VC];
+#pragma unroll
+  for (int k = 0; k < VC; ++k) { best[k] = -INFINITY; bidx[k] = 0; }
+  for (int kh = 0; kh < KS; ++kh) {
+    const int h = p * stride - pad + kh;
+    if (h < 0 || h >= H) continue;
+    for (int kw = 0; kw < KS; ++kw) {
+      const int w = q * stride - pad + kw;
+      if (w < 0 || w >= W) continue;
+      float v[VC];
+      load_group<T, VC>(x + (((int64_t)n * H + h) * W + w) * C + c0, v);
+      const uint8_t wi = (uint8_t)(kh * KS + kw);
+#pragma unroll
+      for (int k = 0; k < VC; ++k)
+        if (v[k] > best[k]) { best[k] = v[k]; bidx[k] = wi; }
+    }
+  }
+  const int64_t o = (((int64_t)n * P + p) * Q + q) * C + c0;
+  store_group<T, VC>(out + o, best);
+#pragma unroll
+  for (int k = 0; k < VC; ++k) idx[o + k] = bidx[k];
+}
+
+template <typename T, int VC>
+__global__ void maxpool_nhwc_bwd_vec_kernel(
+    const T* __restrict__ dout, const uint8_t* __restrict__ idx,
+    T* __restrict__ dx, int C, int H, int W, int P, int Q, int KS, int stride,
+    int pad, int64_t total_chunks) {  // total = N*H*W*(C/VC)
+  const int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= total_chunks) return;
+  const int nc = C / VC;
+  const int c0 = (int)(i % nc) * VC;
+  int64_t r = i / nc;
+  const int w = r % W; r /= W;
+  const int h = r % H; r /= H;
+  const int n = (int)r;
+  float acc[VC];
+#pragma unroll
+  for (int k = 0; k < VC; ++k) acc[k] = 0.f;
+  const int pmin = max(0, (h + pad - KS + stride) / stride);
+  const int pmax = min(P - 1, (h + pad) / stride);
+  const int qmin = max(0, (w + pad - KS + stride) / stride);
+  const int qmax = min(Q - 1, (w + pad) / stride);
+  for (int p = pmin; p <= pmax; ++p) {
+    const int kh = h - (p * stride - pad);
+    if (kh < 0 || kh >= KS) continue;
+    for (int q = qmin; q <= qmax; ++q) {
+      const int kw = w - (q * stride - pad);
+      if (kw < 0 || kw >= KS) continue;
+      const int64_t o = (((int64_t)n * P + p) * Q + q) * C + c0;
+      const uint8_t wi = (uint8_t)(kh * KS + kw);
+      float dv[VC];
+      load_group<T, VC>(dout + o, dv);
+#pragma unroll
+      for (int k = 0; k < VC; ++k)
+        if (idx[o + k] == wi) acc[k] += dv[k];
+    }
+  }
+  store_group<T, VC>(dx + (((int64_t)n * H + h) * W + w) * C + c0, acc);
 }
 
 template <typename T>
@@ -1853,19 +1974,23 @@ inline int64_t nhwc_reduce_blocks(int64_t M, int rows_per_iter, int zslices) {
   return std::min(std::max<int64_t>(want, 1), cap);
 }
 
+// parts = domain branches batched into one launch (grid.y); x is the full
+// (parts*B, H, W, C) NHWC storage, M = per-part positions; acc/mean/cov are
+// stacked [parts, ...].
 void whiten_stats_cl(Tensor x, Tensor acc, Tensor mean, Tensor cov, int64_t g,
-                     int64_t C, int64_t M) {
+                     int64_t C, int64_t M, int64_t parts) {
   const int zslices = (C + 255) / 256;
   DISPATCH_FT(x, "whiten_stats_cl", [&] {
     DWT_SWITCH_G(g, {
       const int GW = (C < 256 ? C : 256) / G;
-      dim3 grid(nhwc_reduce_blocks(M, 256 / GW, zslices), 1, zslices);
+      dim3 grid(nhwc_reduce_blocks(M, 256 / GW, zslices * parts), parts,
+                zslices);
       hipLaunchKernelGGL((dwt::whiten_stats_nhwc_kernel<scalar_t, G>), grid,
                          dim3(256), 0, cur_stream(), x.data_ptr<scalar_t>(),
                          acc.data_ptr<float>(), (int)C, M);
     });
   });
-  const int n_groups = C / g;
+  const int n_groups = parts * C / g;
   const int fin_threads = std::min<int64_t>(g * g, 1024);
   hipLaunchKernelGGL(dwt::whiten_stats_final_kernel, dim3(n_groups),
                      dim3(fin_threads), 0, cur_stream(), acc.data_ptr<float>(),
@@ -1881,12 +2006,12 @@ inline int64_t nhwc_elem_blocks(int64_t M, int rows_per_iter, int zslices) {
 
 void whiten_apply_cl(Tensor x, Tensor mean, Tensor W, Tensor gamma, Tensor beta,
                      Tensor out, int64_t g, int64_t C, int64_t M, bool relu,
-                     bool has_affine) {
+                     bool has_affine, int64_t parts) {
   const int zslices = (C + 255) / 256;
   DISPATCH_FT(x, "whiten_apply_cl", [&] {
     DWT_SWITCH_G(g, {
       const int GW = (C < 256 ? C : 256) / G;
-      dim3 grid(nhwc_elem_blocks(M, 256 / GW, zslices), 1, zslices);
+      dim3 grid(nhwc_elem_blocks(M, 256 / GW, zslices * parts), parts, zslices);
       hipLaunchKernelGGL((dwt::whiten_apply_nhwc_kernel<scalar_t, G>), grid,
                          dim3(256), 0, cur_stream(), x.data_ptr<scalar_t>(),
                          mean.data_ptr<float>(), W.data_ptr<float>(),
@@ -1901,12 +2026,13 @@ void whiten_apply_cl(Tensor x, Tensor mean, Tensor W, Tensor gamma, Tensor beta,
 void whiten_bwd_reduce_cl(Tensor x, Tensor dout, Tensor out, Tensor mean,
                           Tensor W, Tensor gamma, Tensor dWacc, Tensor dgb,
                           int64_t g, int64_t C, int64_t M, bool relu,
-                          bool has_affine) {
+                          bool has_affine, int64_t parts) {
   const int zslices = (C + 255) / 256;
   DISPATCH_FT(x, "whiten_bwd_reduce_cl", [&] {
     DWT_SWITCH_G(g, {
       const int GW = (C < 256 ? C : 256) / G;
-      dim3 grid(nhwc_reduce_blocks(M, 256 / GW, zslices), 1, zslices);
+      dim3 grid(nhwc_reduce_blocks(M, 256 / GW, zslices * parts), parts,
+                zslices);
       hipLaunchKernelGGL((dwt::whiten_bwd_reduce_nhwc_kernel<scalar_t, G>),
                          grid, dim3(256), 0, cur_stream(),
                          x.data_ptr<scalar_t>(), dout.data_ptr<scalar_t>(),
@@ -1922,12 +2048,12 @@ void whiten_bwd_reduce_cl(Tensor x, Tensor dout, Tensor out, Tensor mean,
 void whiten_bwd_apply_cl(Tensor x, Tensor dout, Tensor out, Tensor mean,
                          Tensor W, Tensor gamma, Tensor S, Tensor corr,
                          Tensor dx, int64_t g, int64_t C, int64_t M, bool relu,
-                         bool has_affine, bool train_stats) {
+                         bool has_affine, bool train_stats, int64_t parts) {
   const int zslices = (C + 255) / 256;
   DISPATCH_FT(x, "whiten_bwd_apply_cl", [&] {
     DWT_SWITCH_G(g, {
       const int GW = (C < 256 ? C : 256) / G;
-      dim3 grid(nhwc_elem_blocks(M, 256 / GW, zslices), 1, zslices);
+      dim3 grid(nhwc_elem_blocks(M, 256 / GW, zslices * parts), parts, zslices);
       hipLaunchKernelGGL((dwt::whiten_bwd_apply_nhwc_kernel<scalar_t, G>),
                          grid, dim3(256), 0, cur_stream(),
                          x.data_ptr<scalar_t>(),
@@ -1942,31 +2068,33 @@ void whiten_bwd_apply_cl(Tensor x, Tensor dout, Tensor out, Tensor mean,
 }
 
 void bn_stats_cl(Tensor x, Tensor acc, Tensor mean, Tensor istd, Tensor var_unb,
-                 int64_t C, int64_t M, double eps) {
+                 int64_t C, int64_t M, double eps, int64_t parts) {
   const int zslices = (C + 1023) / 1024;
   DISPATCH_FT(x, "bn_stats_cl", [&] {
     const int NCH = (C < 1024 ? C : 1024) / 4;
-    dim3 grid(nhwc_reduce_blocks(M, 256 / NCH > 0 ? 256 / NCH : 1, zslices), 1,
-              zslices);
+    dim3 grid(nhwc_reduce_blocks(M, 256 / NCH > 0 ? 256 / NCH : 1,
+                                 zslices * parts), parts, zslices);
     hipLaunchKernelGGL((dwt::bn_stats_nhwc_kernel<scalar_t>), grid, dim3(256),
                        0, cur_stream(), x.data_ptr<scalar_t>(),
                        acc.data_ptr<float>(), (int)C, M);
   });
   const float unb = M > 1 ? (float)M / (float)(M - 1) : 1.f;
-  hipLaunchKernelGGL(dwt::bn_stats_final_kernel, dim3((C + 255) / 256),
+  const int total = parts * C;
+  hipLaunchKernelGGL(dwt::bn_stats_final_kernel, dim3((total + 255) / 256),
                      dim3(256), 0, cur_stream(), acc.data_ptr<float>(),
                      mean.data_ptr<float>(), istd.data_ptr<float>(),
-                     var_unb.data_ptr<float>(), (int)C, 1.0f / (float)M, unb,
-                     (float)eps);
+                     var_unb.data_ptr<float>(), total, (int)C,
+                     1.0f / (float)M, unb, (float)eps);
 }
 
 void bn_apply_cl(Tensor x, Tensor mean, Tensor istd, Tensor gamma, Tensor beta,
-                 Tensor out, int64_t C, int64_t M, bool relu, bool has_affine) {
+                 Tensor out, int64_t C, int64_t M, bool relu, bool has_affine,
+                 int64_t parts) {
   const int zslices = (C + 1023) / 1024;
   DISPATCH_FT(x, "bn_apply_cl", [&] {
     const int NCH = (C < 1024 ? C : 1024) / 4;
     const int rpi = std::max(256 / NCH, 1);
-    dim3 grid(nhwc_elem_blocks(M, rpi, zslices), 1, zslices);
+    dim3 grid(nhwc_elem_blocks(M, rpi, zslices * parts), parts, zslices);
     hipLaunchKernelGGL((dwt::bn_apply_nhwc_kernel<scalar_t>), grid,
                        dim3(256), 0, cur_stream(), x.data_ptr<scalar_t>(),
                        mean.data_ptr<float>(), istd.data_ptr<float>(),
@@ -1979,12 +2107,12 @@ void bn_apply_cl(Tensor x, Tensor mean, Tensor istd, Tensor gamma, Tensor beta,
 
 void bn_bwd_reduce_cl(Tensor x, Tensor dout, Tensor out, Tensor mean,
                       Tensor istd, Tensor sums, int64_t C, int64_t M,
-                      bool relu) {
+                      bool relu, int64_t parts) {
   const int zslices = (C + 1023) / 1024;
   DISPATCH_FT(x, "bn_bwd_reduce_cl", [&] {
     const int NCH = (C < 1024 ? C : 1024) / 4;
-    dim3 grid(nhwc_reduce_blocks(M, 256 / NCH > 0 ? 256 / NCH : 1, zslices), 1,
-              zslices);
+    dim3 grid(nhwc_reduce_blocks(M, 256 / NCH > 0 ? 256 / NCH : 1,
+                                 zslices * parts), parts, zslices);
     hipLaunchKernelGGL((dwt::bn_bwd_reduce_nhwc_kernel<scalar_t>), grid,
                        dim3(256), 0, cur_stream(), x.data_ptr<scalar_t>(),
                        dout.data_ptr<scalar_t>(), out.data_ptr<scalar_t>(),
@@ -1996,12 +2124,12 @@ void bn_bwd_reduce_cl(Tensor x, Tensor dout, Tensor out, Tensor mean,
 void bn_bwd_apply_cl(Tensor x, Tensor dout, Tensor out, Tensor mean,
                      Tensor istd, Tensor gamma, Tensor sums, Tensor dx,
                      int64_t C, int64_t M, bool relu, bool has_affine,
-                     bool use_batch) {
+                     bool use_batch, int64_t parts) {
   const int zslices = (C + 1023) / 1024;
   DISPATCH_FT(x, "bn_bwd_apply_cl", [&] {
     const int NCH = (C < 1024 ? C : 1024) / 4;
     const int rpi = std::max(256 / NCH, 1);
-    dim3 grid(nhwc_elem_blocks(M, rpi, zslices), 1, zslices);
+    dim3 grid(nhwc_elem_blocks(M, rpi, zslices * parts), parts, zslices);
     hipLaunchKernelGGL((dwt::bn_bwd_apply_nhwc_kernel<scalar_t>), grid,
                        dim3(256), 0, cur_stream(), x.data_ptr<scalar_t>(),
                        dout.data_ptr<scalar_t>(), out.data_ptr<scalar_t>(),
@@ -2196,8 +2324,8 @@ void bn_stats(Tensor x, Tensor acc, Tensor mean, Tensor istd, Tensor var_unb,
   hipLaunchKernelGGL(dwt::bn_stats_final_kernel,
                      dim3((C + 255) / 256), dim3(256), 0, cur_stream(),
                      acc.data_ptr<float>(), mean.data_ptr<float>(),
-                     istd.data_ptr<float>(), var_unb.data_ptr<float>(), C,
-                     1.0f / (float)M, unb, (float)eps);
+                     istd.data_ptr<float>(), var_unb.data_ptr<float>(), (int)C,
+                     (int)C, 1.0f / (float)M, unb, (float)eps);
 }
 
 void bn_apply(Tensor x, Tensor mean, Tensor istd, Tensor gamma, Tensor beta,
@@ -2281,12 +2409,23 @@ void maxpool_cl_fwd(Tensor x, Tensor out, Tensor idx, int64_t C, int64_t H,
                     int64_t stride, int64_t pad) {
   const int64_t total = out.numel();
   DISPATCH_FT(x, "maxpool_cl_fwd", [&] {
-    hipLaunchKernelGGL((dwt::maxpool_nhwc_fwd_kernel<scalar_t>),
-                       dim3(elementwise_blocks(total, 256)), dim3(256), 0,
-                       cur_stream(), x.data_ptr<scalar_t>(),
-                       out.data_ptr<scalar_t>(), idx.data_ptr<uint8_t>(),
-                       (int)C, (int)H, (int)W, (int)P, (int)Q, (int)KS,
-                       (int)stride, (int)pad, total);
+    constexpr int VC = dwt::VecTraits<scalar_t>::W;
+    if (C % VC == 0) {
+      const int64_t chunks = total / VC;
+      hipLaunchKernelGGL((dwt::maxpool_nhwc_fwd_vec_kernel<scalar_t, VC>),
+                         dim3(elementwise_blocks(chunks, 256)), dim3(256), 0,
+                         cur_stream(), x.data_ptr<scalar_t>(),
+                         out.data_ptr<scalar_t>(), idx.data_ptr<uint8_t>(),
+                         (int)C, (int)H, (int)W, (int)P, (int)Q, (int)KS,
+                         (int)stride, (int)pad, chunks);
+    } else {
+      hipLaunchKernelGGL((dwt::maxpool_nhwc_fwd_kernel<scalar_t>),
+                         dim3(elementwise_blocks(total, 256)), dim3(256), 0,
+                         cur_stream(), x.data_ptr<scalar_t>(),
+                         out.data_ptr<scalar_t>(), idx.data_ptr<uint8_t>(),
+                         (int)C, (int)H, (int)W, (int)P, (int)Q, (int)KS,
+                         (int)stride, (int)pad, total);
+    }
   });
 }
 
@@ -2295,12 +2434,23 @@ void maxpool_cl_bwd(Tensor dout, Tensor idx, Tensor dx, int64_t C, int64_t H,
                     int64_t stride, int64_t pad) {
   const int64_t total_in = dx.numel();
   DISPATCH_FT(dout, "maxpool_cl_bwd", [&] {
-    hipLaunchKernelGGL((dwt::maxpool_nhwc_bwd_kernel<scalar_t>),
-                       dim3(elementwise_blocks(total_in, 256)), dim3(256), 0,
-                       cur_stream(), dout.data_ptr<scalar_t>(),
-                       idx.data_ptr<uint8_t>(), dx.data_ptr<scalar_t>(),
-                       (int)C, (int)H, (int)W, (int)P, (int)Q, (int)KS,
-                       (int)stride, (int)pad, total_in);
+    constexpr int VC = dwt::VecTraits<scalar_t>::W;
+    if (C % VC == 0) {
+      const int64_t chunks = total_in / VC;
+      hipLaunchKernelGGL((dwt::maxpool_nhwc_bwd_vec_kernel<scalar_t, VC>),
+                         dim3(elementwise_blocks(chunks, 256)), dim3(256), 0,
+                         cur_stream(), dout.data_ptr<scalar_t>(),
+                         idx.data_ptr<uint8_t>(), dx.data_ptr<scalar_t>(),
+                         (int)C, (int)H, (int)W, (int)P, (int)Q, (int)KS,
+                         (int)stride, (int)pad, chunks);
+    } else {
+      hipLaunchKernelGGL((dwt::maxpool_nhwc_bwd_kernel<scalar_t>),
+                         dim3(elementwise_blocks(total_in, 256)), dim3(256), 0,
+                         cur_stream(), dout.data_ptr<scalar_t>(),
+                         idx.data_ptr<uint8_t>(), dx.data_ptr<scalar_t>(),
+                         (int)C, (int)H, (int)W, (int)P, (int)Q, (int)KS,
+                         (int)stride, (int)pad, total_in);
+    }
   });
 }
 
