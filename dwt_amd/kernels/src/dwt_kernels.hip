@@ -2636,10 +2636,10 @@ void mfma_conv2d_dgrad(torch::Tensor dy, torch::Tensor wd, torch::Tensor dx,
                        int64_t N, int64_t H, int64_t W, int64_t Cin,
                        int64_t P, int64_t Q, int64_t Cdy, int64_t KH,
                        int64_t KW, int64_t stride, int64_t pad);
-void mfma_conv2d_wgrad(torch::Tensor dyT, torch::Tensor x, torch::Tensor dw,
-                       int64_t N, int64_t H, int64_t W, int64_t Cin,
-                       int64_t P, int64_t Q, int64_t Cout, int64_t KH,
-                       int64_t KW, int64_t stride, int64_t pad);
+void mfma_conv2d_wgrad(torch::Tensor dy, torch::Tensor x, torch::Tensor dw,
+                       torch::Tensor ws, int64_t N, int64_t H, int64_t W,
+                       int64_t Cin, int64_t P, int64_t Q, int64_t Cout,
+                       int64_t KH, int64_t KW, int64_t stride, int64_t pad);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("mfma_gemm", &mfma_gemm);

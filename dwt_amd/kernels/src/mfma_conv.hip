@@ -50,6 +50,7 @@ struct ConvParams {
   int64_t M;             // N*P*Q (fwd) / N*H*W (dgrad)
   int K;                 // KH*KW*Cin (fwd) / KH*KW*Cdy (dgrad)
   int Cdy;               // dgrad: channels of dy
+  int cShift = -1;       // log2 of the K-fastest channel dim if pow2
 };
 
 DEV_INLINE float bf16_to_f(unsigned short u) {
@@ -76,39 +77,88 @@ struct StageRegs {
 // A-operand gather modes
 enum AMode { A_DENSE = 0, A_CONV = 1, A_DGRAD = 2 };
 
+// Per-tile row cache: each thread's 4 chunk rows keep the same output
+// position m across the whole K loop, so the m -> (n, p, q) decode (two
+// 64-bit div/mod chains) runs ONCE per tile instead of once per k-step.
+struct RowCache {
+  int h0[4], w0[4];     // A_CONV: p*stride-pad / q*stride-pad; A_DGRAD: h+pad
+  int64_t base[4];      // image base offset (elements)
+  bool valid[4];
+};
+
+template <int MODE>
+DEV_INLINE RowCache make_rows(const ConvParams& cp, int64_t m0) {
+  RowCache rc;
+  const int t = threadIdx.x;
+#pragma unroll
+  for (int c = 0; c < 4; ++c) {
+    const int idx = t + c * THREADS;
+    const int row = idx / (BK / 8);
+    const int64_t m = m0 + row;
+    rc.valid[c] = m < cp.M;
+    const int64_t mm = rc.valid[c] ? m : 0;
+    if (MODE == A_CONV) {
+      const int q = (int)(mm % cp.Q);
+      const int64_t np = mm / cp.Q;
+      const int p = (int)(np % cp.P);
+      const int n = (int)(np / cp.P);
+      rc.h0[c] = p * cp.stride - cp.pad;
+      rc.w0[c] = q * cp.stride - cp.pad;
+      rc.base[c] = (int64_t)n * cp.H * cp.W * cp.Cin;
+    } else if (MODE == A_DGRAD) {
+      const int wi = (int)(mm % cp.W);
+      const int64_t nh = mm / cp.W;
+      const int hi = (int)(nh % cp.H);
+      const int n = (int)(nh / cp.H);
+      rc.h0[c] = hi + cp.pad;
+      rc.w0[c] = wi + cp.pad;
+      rc.base[c] = (int64_t)n * cp.P * cp.Q * cp.Cdy;
+    } else {
+      rc.h0[c] = 0; rc.w0[c] = 0;
+      rc.base[c] = mm * cp.K;
+    }
+  }
+  return rc;
+}
+
+// cShift >= 0 when the K-fastest channel count (Cin fwd / Cdy dgrad) is a
+// power of two: replaces the per-chunk div/mod with shift/mask.
+DEV_INLINE void split_kc(int kg, int cdim, int cshift, int& rs, int& cc) {
+  if (cshift >= 0) {
+    cc = kg & (cdim - 1);
+    rs = kg >> cshift;
+  } else {
+    cc = kg % cdim;
+    rs = kg / cdim;
+  }
+}
+
 template <int MODE>
 DEV_INLINE void load_a(const bf16* __restrict__ x, const ConvParams& cp,
-                       int64_t m0, int k0, StageRegs& rg) {
+                       const RowCache& rc, int k0, StageRegs& rg) {
   const int t = threadIdx.x;
 #pragma unroll
   for (int c = 0; c < 4; ++c) {
     const int idx = t + c * THREADS;          // chunk index in tile
-    const int row = idx / (BK / 8);           // 0..127
     const int kc = (idx % (BK / 8)) * 8;      // chunk k offset in tile
-    const int64_t m = m0 + row;
     const int kg = k0 + kc;
-    if (m >= cp.M || kg >= cp.K) {
+    if (!rc.valid[c] || kg >= cp.K) {
       rg.c[c] = make_uint4(0, 0, 0, 0);
       continue;
     }
     if (MODE == A_DENSE) {
-      rg.c[c] = *reinterpret_cast<const uint4*>(x + m * cp.K + kg);
+      rg.c[c] = *reinterpret_cast<const uint4*>(x + rc.base[c] + kg);
       continue;
     }
     if (MODE == A_DGRAD) {
-      // x here is dy (N,P,Q,Cdy) raw NHWC; m -> input position (n,h,w);
-      // k -> (r, s, co) with co fastest.  dx[n,h,w,ci] needs
-      // dy[n, (h+pad-r)/stride, (w+pad-s)/stride, co] when divisible.
-      const int wi = (int)(m % cp.W);
-      const int64_t nh = m / cp.W;
-      const int hi = (int)(nh % cp.H);
-      const int n = (int)(nh / cp.H);
-      const int co = kg % cp.Cdy;
-      const int rs = kg / cp.Cdy;
+      // x here is dy (N,P,Q,Cdy) raw NHWC; k -> (r, s, co), co fastest.
+      // dx[n,h,w,ci] needs dy[n, (h+pad-r)/stride, (w+pad-s)/stride, co].
+      int rs, co;
+      split_kc(kg, cp.Cdy, cp.cShift, rs, co);
       const int sx = rs % cp.KW;
       const int r = rs / cp.KW;
-      const int hp = hi + cp.pad - r;
-      const int wp = wi + cp.pad - sx;
+      const int hp = rc.h0[c] - r;
+      const int wp = rc.w0[c] - sx;
       bool ok = hp >= 0 && wp >= 0 && hp % cp.stride == 0 && wp % cp.stride == 0;
       const int pp = hp / cp.stride, qq = wp / cp.stride;
       ok = ok && pp < cp.P && qq < cp.Q;
@@ -116,7 +166,7 @@ DEV_INLINE void load_a(const bf16* __restrict__ x, const ConvParams& cp,
         rg.c[c] = make_uint4(0, 0, 0, 0);
       } else if (cp.Cdy % 8 == 0) {
         rg.c[c] = *reinterpret_cast<const uint4*>(
-            x + (((int64_t)n * cp.P + pp) * cp.Q + qq) * cp.Cdy + co);
+            x + rc.base[c] + ((int64_t)pp * cp.Q + qq) * cp.Cdy + co);
       } else {
         unsigned short tmp[8];
 #pragma unroll
@@ -125,29 +175,25 @@ DEV_INLINE void load_a(const bf16* __restrict__ x, const ConvParams& cp,
           unsigned short v = 0;
           if (k < cp.K && k / cp.Cdy == rs)
             v = *reinterpret_cast<const unsigned short*>(
-                x + (((int64_t)n * cp.P + pp) * cp.Q + qq) * cp.Cdy + k % cp.Cdy);
+                x + rc.base[c] + ((int64_t)pp * cp.Q + qq) * cp.Cdy + k % cp.Cdy);
           tmp[j] = v;
         }
         rg.c[c] = *reinterpret_cast<const uint4*>(tmp);
       }
       continue;
     }
-    const int q = (int)(m % cp.Q);
-    const int64_t np = m / cp.Q;
-    const int p = (int)(np % cp.P);
-    const int n = (int)(np / cp.P);
     if (cp.Cin % 8 == 0) {
-      const int ci = kg % cp.Cin;
-      const int rs = kg / cp.Cin;
+      int rs, ci;
+      split_kc(kg, cp.Cin, cp.cShift, rs, ci);
       const int s = rs % cp.KW;
       const int r = rs / cp.KW;
-      const int h = p * cp.stride - cp.pad + r;
-      const int w = q * cp.stride - cp.pad + s;
+      const int h = rc.h0[c] + r;
+      const int w = rc.w0[c] + s;
       if (h < 0 || h >= cp.H || w < 0 || w >= cp.W) {
         rg.c[c] = make_uint4(0, 0, 0, 0);
       } else {
         rg.c[c] = *reinterpret_cast<const uint4*>(
-            x + (((int64_t)n * cp.H + h) * cp.W + w) * cp.Cin + ci);
+            x + rc.base[c] + ((int64_t)h * cp.W + w) * cp.Cin + ci);
       }
     } else {
       unsigned short tmp[8];
@@ -160,11 +206,11 @@ DEV_INLINE void load_a(const bf16* __restrict__ x, const ConvParams& cp,
           const int rs = k / cp.Cin;
           const int s = rs % cp.KW;
           const int r = rs / cp.KW;
-          const int h = p * cp.stride - cp.pad + r;
-          const int w = q * cp.stride - cp.pad + s;
+          const int h = rc.h0[c] + r;
+          const int w = rc.w0[c] + s;
           if (h >= 0 && h < cp.H && w >= 0 && w < cp.W)
             v = *reinterpret_cast<const unsigned short*>(
-                x + (((int64_t)n * cp.H + h) * cp.W + w) * cp.Cin + ci);
+                x + rc.base[c] + ((int64_t)h * cp.W + w) * cp.Cin + ci);
         }
         tmp[j] = v;
       }
@@ -199,51 +245,6 @@ DEV_INLINE void load_b(const bf16* __restrict__ wgt, int ncols, int K,
   }
 }
 
-// wgrad B-operand gather (see mfma_conv2d_wgrad): row n' = (r*KW+s)*Cin+ci,
-// k = flattened (n, p, q) output position of the forward conv; per-element
-// decode (k-chunks cross q rows), zeros outside bounds or past the real NPQ
-// (K is padded to a multiple of 8 for the dense A loads).
-DEV_INLINE void load_b_wgrad(const bf16* __restrict__ x, const ConvParams& cp,
-                             int n0, int k0, StageRegs& rg) {
-  const int t = threadIdx.x;
-  const int npq = cp.Cdy;  // real (unpadded) N*P*Q
-#pragma unroll
-  for (int c = 0; c < 4; ++c) {
-    const int idx = t + c * THREADS;
-    const int row = idx / (BK / 8);
-    const int kc = (idx % (BK / 8)) * 8;
-    const int nrow = n0 + row;
-    const int kg = k0 + kc;
-    if (nrow >= cp.Cout || kg >= cp.K) {  // cp.Cout = R*S*Cin columns of dw
-      rg.c[c] = make_uint4(0, 0, 0, 0);
-      continue;
-    }
-    const int ci = nrow % cp.Cin;
-    const int rs = nrow / cp.Cin;
-    const int sx = rs % cp.KW;
-    const int r = rs / cp.KW;
-    unsigned short tmp[8];
-#pragma unroll
-    for (int j = 0; j < 8; ++j) {
-      const int k = kg + j;
-      unsigned short v = 0;
-      if (k < npq) {
-        const int q = k % cp.Q;
-        const int np = k / cp.Q;
-        const int pp = np % cp.P;
-        const int n = np / cp.P;
-        const int h = pp * cp.stride - cp.pad + r;
-        const int w = q * cp.stride - cp.pad + sx;
-        if (h >= 0 && h < cp.H && w >= 0 && w < cp.W)
-          v = *reinterpret_cast<const unsigned short*>(
-              x + (((int64_t)n * cp.H + h) * cp.W + w) * cp.Cin + ci);
-      }
-      tmp[j] = v;
-    }
-    rg.c[c] = *reinterpret_cast<const uint4*>(tmp);
-  }
-}
-
 DEV_INLINE void write_tile(const StageRegs& rg, unsigned short* lds) {
   const int t = threadIdx.x;
 #pragma unroll
@@ -260,8 +261,7 @@ DEV_INLINE void write_tile(const StageRegs& rg, unsigned short* lds) {
 // single-buffer loop wins for short-K and for the implicit gather path
 // (within-shape A/B on the R50 shapes).  SWZ: XCD-aware bijective block
 // remap (guide T1) — only when the grid has several N-tiles to share.
-template <int MODE, bool RELU, bool HAS_BIAS, bool PIPE, bool SWZ,
-          bool WGRAD_B = false>
+template <int MODE, bool RELU, bool HAS_BIAS, bool PIPE, bool SWZ>
 __global__ __launch_bounds__(THREADS, 2) void conv_implicit_gemm_kernel(
     const bf16* __restrict__ x, const bf16* __restrict__ wgt,
     const float* __restrict__ bias, bf16* __restrict__ out, ConvParams cp,
@@ -296,10 +296,10 @@ __global__ __launch_bounds__(THREADS, 2) void conv_implicit_gemm_kernel(
   const int nk = (cp.K + BK - 1) / BK;
 
   StageRegs ra, rb;
+  const RowCache rc = make_rows<MODE>(cp, m0);
   if (PIPE) {
-    load_a<MODE>(x, cp, m0, 0, ra);
-    if (WGRAD_B) load_b_wgrad(wgt, cp, n0, 0, rb);
-    else load_b(wgt, cp.Cout, cp.K, n0, 0, rb);
+    load_a<MODE>(x, cp, rc, 0, ra);
+    load_b(wgt, cp.Cout, cp.K, n0, 0, rb);
     write_tile(ra, lds_a[0]);
     write_tile(rb, lds_b[0]);
     __syncthreads();
@@ -311,14 +311,12 @@ __global__ __launch_bounds__(THREADS, 2) void conv_implicit_gemm_kernel(
       // issue next tile's global loads now — they stay in flight under the
       // MFMA phase and are only waited for at the ds_write below
       if (t + 1 < nk) {
-        load_a<MODE>(x, cp, m0, (t + 1) * BK, ra);
-        if (WGRAD_B) load_b_wgrad(wgt, cp, n0, (t + 1) * BK, rb);
-        else load_b(wgt, cp.Cout, cp.K, n0, (t + 1) * BK, rb);
+        load_a<MODE>(x, cp, rc, (t + 1) * BK, ra);
+        load_b(wgt, cp.Cout, cp.K, n0, (t + 1) * BK, rb);
       }
     } else {
-      load_a<MODE>(x, cp, m0, t * BK, ra);
-      if (WGRAD_B) load_b_wgrad(wgt, cp, n0, t * BK, rb);
-      else load_b(wgt, cp.Cout, cp.K, n0, t * BK, rb);
+      load_a<MODE>(x, cp, rc, t * BK, ra);
+      load_b(wgt, cp.Cout, cp.K, n0, t * BK, rb);
       write_tile(ra, lds_a[0]);
       write_tile(rb, lds_b[0]);
       __syncthreads();
@@ -376,6 +374,177 @@ __global__ __launch_bounds__(THREADS, 2) void conv_implicit_gemm_kernel(
   }
 }
 
+// ===========================================================================
+// wgrad v2: dw[co][(r,s,ci)] = sum_{k=npq} dy[k][co] * xpatch[k][(r,s,ci)]
+//
+// The K dimension is the (huge) output-position axis, so the kernel is
+// SPLIT-K: the grid is (mtiles * ntiles * splitk) and each block accumulates
+// its K-slab into an fp32 workspace with atomicAdd; a tiny convert kernel
+// writes the bf16 dw afterwards.  Both operands are K-major with stride C
+// in memory, so tiles are loaded coalesced in (k, c) orientation and
+// TRANSPOSED through LDS into the (c, k) fragment layout the MFMA wants.
+// 64x64 tile, BK=64, 4 waves each doing a 32x32 quadrant.
+//
+// B (x patches): when Cin % 64 == 0 every 64-column N-tile lies inside one
+// (r, s) tap, so the gather is one shifted NHWC base + coalesced 16-B
+// chunks; otherwise (stem Cin=3) a scalar per-element gather.
+// ===========================================================================
+
+
+
+constexpr int WBM = 64, WBN = 64, WBK = 64;
+constexpr int WPITCH = WBK + 8;  // halves; +16 B keeps the strided
+                                 // transpose writes off one bank
+
+template <bool ALIGNED_B>
+__global__ __launch_bounds__(256, 4) void wgrad_splitk_kernel(
+    const bf16* __restrict__ dy, const bf16* __restrict__ x,
+    float* __restrict__ ws, ConvParams cp, int mtiles, int ntiles,
+    int splitk, int64_t klen) {
+  __shared__ unsigned short lds_a[WBM * WPITCH];  // [co][k]
+  __shared__ unsigned short lds_b[WBN * WPITCH];  // [rsci][k]
+
+  const int bid = blockIdx.x;
+  const int tile = bid / splitk;
+  const int kchunk = bid % splitk;
+  const int m0 = (tile / ntiles) * WBM;   // co origin
+  const int n0 = (tile % ntiles) * WBN;   // rsci origin
+  const int64_t ks = (int64_t)kchunk * klen;
+  const int64_t ke = (ks + klen < cp.M) ? ks + klen : cp.M;  // cp.M = real NPQ
+
+  const int t = threadIdx.x;
+  const int wave = t / 64;
+  const int lane = t % 64;
+  const int wm = (wave / 2) * 32;
+  const int wn = (wave % 2) * 32;
+
+  // B-tile tap decode (single (r,s) per tile when ALIGNED_B)
+  int rB = 0, sB = 0, ciB = 0;
+  if (ALIGNED_B) {
+    const int rs = n0 / cp.Cin;
+    sB = rs % cp.KW;
+    rB = rs / cp.KW;
+    ciB = n0 % cp.Cin;
+  }
+
+  floatx4 acc[2][2];
+#pragma unroll
+  for (int i = 0; i < 2; ++i)
+#pragma unroll
+    for (int j = 0; j < 2; ++j) acc[i][j] = (floatx4){0.f, 0.f, 0.f, 0.f};
+
+  const int frag_row = lane % 16;
+  const int frag_koff = (lane / 16) * 8;
+
+  for (int64_t k0 = ks; k0 < ke; k0 += WBK) {
+    // ---- load dy tile [WBK k][WBM co] coalesced, transpose into lds_a ----
+    // thread t covers (krow = t/8, co chunk = (t%8)*8), two passes of 32 rows
+#pragma unroll
+    for (int pass = 0; pass < 2; ++pass) {
+      const int krow = (t / 8) + pass * 32;
+      const int co = (t % 8) * 8;
+      const int64_t k = k0 + krow;
+      unsigned short v[8];
+      if (k < ke && m0 + co < cp.N) {  // cp.N = Cout here
+        const uint4 u = *reinterpret_cast<const uint4*>(
+            dy + k * cp.N + m0 + co);
+        *reinterpret_cast<uint4*>(v) = u;
+      } else {
+#pragma unroll
+        for (int j = 0; j < 8; ++j) v[j] = 0;
+      }
+#pragma unroll
+      for (int j = 0; j < 8; ++j) lds_a[(co + j) * WPITCH + krow] = v[j];
+    }
+    // ---- load x-patch tile [WBK k][WBN rsci], transpose into lds_b ----
+#pragma unroll
+    for (int pass = 0; pass < 2; ++pass) {
+      const int krow = (t / 8) + pass * 32;
+      const int nc = (t % 8) * 8;       // rsci offset within tile
+      const int64_t k = k0 + krow;
+      unsigned short v[8];
+#pragma unroll
+      for (int j = 0; j < 8; ++j) v[j] = 0;
+      if (k < ke) {
+        const int q = (int)(k % cp.Q);
+        const int64_t np = k / cp.Q;
+        const int p = (int)(np % cp.P);
+        const int n = (int)(np / cp.P);
+        if (ALIGNED_B) {
+          const int h = p * cp.stride - cp.pad + rB;
+          const int w = q * cp.stride - cp.pad + sB;
+          if (h >= 0 && h < cp.H && w >= 0 && w < cp.W) {
+            const uint4 u = *reinterpret_cast<const uint4*>(
+                x + (((int64_t)n * cp.H + h) * cp.W + w) * cp.Cin + ciB + nc);
+            *reinterpret_cast<uint4*>(v) = u;
+          }
+        } else {
+#pragma unroll
+          for (int j = 0; j < 8; ++j) {
+            const int nr = n0 + nc + j;
+            if (nr >= cp.Cout) break;   // cp.Cout = KH*KW*Cin columns
+            const int ci = nr % cp.Cin;
+            const int rs = nr / cp.Cin;
+            const int sx = rs % cp.KW;
+            const int r = rs / cp.KW;
+            const int h = p * cp.stride - cp.pad + r;
+            const int w = q * cp.stride - cp.pad + sx;
+            if (h >= 0 && h < cp.H && w >= 0 && w < cp.W)
+              v[j] = *reinterpret_cast<const unsigned short*>(
+                  x + (((int64_t)n * cp.H + h) * cp.W + w) * cp.Cin + ci);
+          }
+        }
+      }
+#pragma unroll
+      for (int j = 0; j < 8; ++j) lds_b[(nc + j) * WPITCH + krow] = v[j];
+    }
+    __syncthreads();
+#pragma unroll
+    for (int kstep = 0; kstep < WBK; kstep += 32) {
+      short8 afrag[2], bfrag[2];
+#pragma unroll
+      for (int i = 0; i < 2; ++i) {
+        afrag[i] = *reinterpret_cast<const short8*>(
+            lds_a + (wm + i * 16 + frag_row) * WPITCH + kstep + frag_koff);
+        bfrag[i] = *reinterpret_cast<const short8*>(
+            lds_b + (wn + i * 16 + frag_row) * WPITCH + kstep + frag_koff);
+      }
+#pragma unroll
+      for (int i = 0; i < 2; ++i)
+#pragma unroll
+        for (int j = 0; j < 2; ++j)
+          acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              afrag[i], bfrag[j], acc[i][j], 0, 0, 0);
+    }
+    __syncthreads();
+  }
+
+  // epilogue: atomicAdd fp32 workspace [Cout][RSC]
+  const int erow = (lane / 16) * 4;
+  const int ecol = lane % 16;
+#pragma unroll
+  for (int i = 0; i < 2; ++i) {
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      const int m = m0 + wm + i * 16 + erow + r;
+      if (m >= cp.N) continue;
+#pragma unroll
+      for (int j = 0; j < 2; ++j) {
+        const int n = n0 + wn + j * 16 + ecol;
+        if (n >= cp.Cout) continue;
+        atomicAdd(&ws[(int64_t)m * cp.Cout + n], acc[i][j][r]);
+      }
+    }
+  }
+}
+
+__global__ void f32_to_bf16_kernel(const float* __restrict__ src,
+                                   bf16* __restrict__ dst, int64_t n) {
+  const int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  if (i < n)
+    *reinterpret_cast<unsigned short*>(dst + i) = f_to_bf16(src[i]);
+}
+
 }  // namespace dwtmm
 
 // ---------------------------------------------------------------------------
@@ -386,6 +555,10 @@ using torch::Tensor;
 
 static inline hipStream_t dwtmm_stream() {
   return at::cuda::getCurrentCUDAStream().stream();
+}
+
+static inline int log2_if_pow2(int v) {
+  return (v > 0 && (v & (v - 1)) == 0) ? __builtin_ctz((unsigned)v) : -1;
 }
 
 // C[M,N] = A[M,K] @ Bt[N,K]^T (+bias +relu), all bf16, fp32 accumulate.
@@ -438,13 +611,14 @@ void mfma_conv2d_fwd(Tensor x, Tensor wgt, Tensor bias, Tensor out,
   cp.KH = KH; cp.KW = KW; cp.stride = stride; cp.pad = pad;
   cp.M = N * P * Q;
   cp.K = KH * KW * Cin;
+  cp.cShift = log2_if_pow2(Cin);
   const int mtiles = (cp.M + dwtmm::BM - 1) / dwtmm::BM;
   const int ntiles = (Cout + dwtmm::BN - 1) / dwtmm::BN;
   const bool gemm_fast = (KH == 1 && KW == 1 && stride == 1 && pad == 0 &&
                           Cin % 8 == 0);
-  // pipelined staging only pays for long-K dense GEMMs (A/B measured);
-  // the implicit gather path keeps the simple loop, no swizzle
-  const bool pipe = gemm_fast && cp.K >= 4 * dwtmm::BK;
+  // pipelined (double-buffered, issue-early) staging for every medium/long-K
+  // shape; the gather path's per-k decode now reuses the per-tile row cache
+  const bool pipe = cp.K >= 4 * dwtmm::BK;
   const bool swz = gemm_fast && ntiles >= 8 && mtiles * ntiles >= 16;
   auto run = [&](auto fastc, auto reluc, auto biasc) {
     auto launch = [&](auto pipec, auto swzc) {
@@ -489,40 +663,74 @@ void mfma_conv2d_dgrad(Tensor dy, Tensor wd, Tensor dx, int64_t N, int64_t H,
   cp.Cdy = Cdy;
   cp.M = N * H * W;
   cp.K = KH * KW * Cdy;
+  cp.cShift = log2_if_pow2(Cdy);
   const int mtiles = (cp.M + dwtmm::BM - 1) / dwtmm::BM;
   const int ntiles = (Cin + dwtmm::BN - 1) / dwtmm::BN;
-  hipLaunchKernelGGL(
-      (dwtmm::conv_implicit_gemm_kernel<dwtmm::A_DGRAD, false, false, false,
-                                        false>),
-      dim3(mtiles * ntiles), dim3(dwtmm::THREADS), 0, dwtmm_stream(),
-      (const c10::BFloat16*)dy.data_ptr(), (const c10::BFloat16*)wd.data_ptr(),
-      nullptr, (c10::BFloat16*)dx.data_ptr(), cp, mtiles, ntiles);
+  // 1x1/stride-1 dgrad IS the dense GEMM dy @ w^T — run it on the
+  // pipelined/swizzled dense path instead of the per-chunk gather decode
+  const bool gemm_fast = (KH == 1 && KW == 1 && stride == 1 && pad == 0 &&
+                          Cdy % 8 == 0);
+  const bool pipe = cp.K >= 4 * dwtmm::BK;
+  const bool swz = gemm_fast && ntiles >= 8 && mtiles * ntiles >= 16;
+  auto launch = [&](auto modec, auto pipec, auto swzc) {
+    hipLaunchKernelGGL(
+        (dwtmm::conv_implicit_gemm_kernel<decltype(modec)::value,
+                                          false, false,
+                                          decltype(pipec)::value,
+                                          decltype(swzc)::value>),
+        dim3(mtiles * ntiles), dim3(dwtmm::THREADS), 0, dwtmm_stream(),
+        (const c10::BFloat16*)dy.data_ptr(), (const c10::BFloat16*)wd.data_ptr(),
+        nullptr, (c10::BFloat16*)dx.data_ptr(), cp, mtiles, ntiles);
+  };
+  using DEN = std::integral_constant<int, dwtmm::A_DENSE>;
+  using DGR = std::integral_constant<int, dwtmm::A_DGRAD>;
+  if (gemm_fast) {
+    if (pipe) { if (swz) launch(DEN{}, std::true_type{}, std::true_type{});
+                else launch(DEN{}, std::true_type{}, std::false_type{}); }
+    else { if (swz) launch(DEN{}, std::false_type{}, std::true_type{});
+           else launch(DEN{}, std::false_type{}, std::false_type{}); }
+  } else {
+    if (pipe) launch(DGR{}, std::true_type{}, std::false_type{});
+    else launch(DGR{}, std::false_type{}, std::false_type{});
+  }
 }
 
 
-// wgrad: dw[co][(r,s,ci)] (= channels_last weight storage) as the GEMM
-//   A[co][k=npq] = dy^T (host-transposed, K padded to %8)
-//   Bt[(r,s,ci)][k] = x patches (gathered in-kernel)
-// cp.M = Cout, cp.Cout = KH*KW*Cin (dw columns), cp.K = padded NPQ,
-// cp.Cdy = real NPQ.
-void mfma_conv2d_wgrad(Tensor dyT, Tensor x, Tensor dw, int64_t N, int64_t H,
-                       int64_t W, int64_t Cin, int64_t P, int64_t Q,
-                       int64_t Cout, int64_t KH, int64_t KW, int64_t stride,
-                       int64_t pad) {
+// wgrad v2 (split-K): dy (N,P,Q,Cout) raw NHWC, x (N,H,W,Cin) raw NHWC,
+// ws fp32 zero-initialized [Cout, KH*KW*Cin], dw bf16 channels_last weight
+// storage (same [co][r][s][ci] flat layout as ws).
+// ConvParams reuse here: cp.N = Cout (dy channel stride), cp.Cout =
+// KH*KW*Cin (dw columns), cp.M = real NPQ.
+void mfma_conv2d_wgrad(Tensor dy, Tensor x, Tensor dw, Tensor ws, int64_t N,
+                       int64_t H, int64_t W, int64_t Cin, int64_t P,
+                       int64_t Q, int64_t Cout, int64_t KH, int64_t KW,
+                       int64_t stride, int64_t pad) {
   dwtmm::ConvParams cp{};
-  cp.N = N; cp.H = H; cp.W = W; cp.Cin = Cin;
+  cp.N = Cout; cp.H = H; cp.W = W; cp.Cin = Cin;
   cp.P = P; cp.Q = Q;
   cp.KH = KH; cp.KW = KW; cp.stride = stride; cp.pad = pad;
-  cp.M = Cout;
+  cp.M = (int64_t)N * P * Q;
   cp.Cout = KH * KW * Cin;
-  cp.K = dyT.size(1);          // padded
-  cp.Cdy = N * P * Q;          // real
-  const int mtiles = (cp.M + dwtmm::BM - 1) / dwtmm::BM;
-  const int ntiles = (cp.Cout + dwtmm::BN - 1) / dwtmm::BN;
-  hipLaunchKernelGGL(
-      (dwtmm::conv_implicit_gemm_kernel<dwtmm::A_DENSE, false, false, false,
-                                        false, true>),
-      dim3(mtiles * ntiles), dim3(dwtmm::THREADS), 0, dwtmm_stream(),
-      (const c10::BFloat16*)dyT.data_ptr(), (const c10::BFloat16*)x.data_ptr(),
-      nullptr, (c10::BFloat16*)dw.data_ptr(), cp, mtiles, ntiles);
+  const int mtiles = (Cout + dwtmm::WBM - 1) / dwtmm::WBM;
+  const int ntiles = (cp.Cout + dwtmm::WBN - 1) / dwtmm::WBN;
+  // split K so the grid lands at >=2048 workgroups (256 CUs, several waves
+  // deep), k-slabs rounded to whole BK tiles
+  int64_t splitk = std::max<int64_t>(1, 2048 / (mtiles * ntiles));
+  const int64_t kt = (cp.M + dwtmm::WBK - 1) / dwtmm::WBK;  // total k-tiles
+  splitk = std::min<int64_t>(splitk, kt);
+  const int64_t klen = ((kt + splitk - 1) / splitk) * dwtmm::WBK;
+  const bool aligned = (Cin % 64 == 0);
+  auto launch = [&](auto ac) {
+    hipLaunchKernelGGL(
+        (dwtmm::wgrad_splitk_kernel<decltype(ac)::value>),
+        dim3(mtiles * ntiles * splitk), dim3(256), 0, dwtmm_stream(),
+        (const c10::BFloat16*)dy.data_ptr(), (const c10::BFloat16*)x.data_ptr(),
+        ws.data_ptr<float>(), cp, mtiles, ntiles, (int)splitk, klen);
+  };
+  if (aligned) launch(std::true_type{});
+  else launch(std::false_type{});
+  const int64_t n = (int64_t)Cout * cp.Cout;
+  hipLaunchKernelGGL(dwtmm::f32_to_bf16_kernel, dim3((n + 255) / 256),
+                     dim3(256), 0, dwtmm_stream(), ws.data_ptr<float>(),
+                     (c10::BFloat16*)dw.data_ptr(), n);
 }

@@ -207,17 +207,19 @@ def _wgrad_patches_reference(x_nhwc, shape, stride, padding, npq_pad=None,
 
 def conv2d_wgrad(dy: torch.Tensor, x: torch.Tensor, weight_shape,
                  stride: int = 1, padding: int = 0) -> torch.Tensor:
-    """dw for an NHWC conv on the MFMA kernel: A = dy^T (host transpose),
-    B = input patches gathered in-kernel; output IS the channels_last
-    weight storage [co][r][s][ci]."""
+    """dw for an NHWC conv on the split-K MFMA wgrad kernel: both operands
+    consumed K-major straight from their NHWC storage (LDS-transposed in
+    kernel), K = N*P*Q split across the grid into an fp32 workspace;
+    output IS the channels_last weight storage [co][r][s][ci]."""
     cout, cin, kh, kw = weight_shape
     n, _, p, q = dy.shape
+    assert cout % 8 == 0, "wgrad kernel needs Cout % 8 == 0"
     dyc = dy.contiguous(memory_format=torch.channels_last)
-    dyt = dyc.permute(0, 2, 3, 1).reshape(n * p * q, cout).t().contiguous()
-    dyt = _pad_k(dyt)
     dw = torch.empty(weight_shape, device=dy.device, dtype=torch.bfloat16,
                      memory_format=torch.channels_last)
+    ws = torch.zeros(cout * kh * kw * cin, device=dy.device,
+                     dtype=torch.float32)
     xc = x.contiguous(memory_format=torch.channels_last)
-    _ext().mfma_conv2d_wgrad(dyt, xc, dw, n, x.shape[2], x.shape[3], cin,
+    _ext().mfma_conv2d_wgrad(dyc, xc, dw, ws, n, x.shape[2], x.shape[3], cin,
                              p, q, cout, kh, kw, stride, padding)
     return dw

@@ -145,6 +145,9 @@ def test_mfma_conv2d_module_trains(dev):
     (4, 64, 28, 28, 64, 1, 1, 0),
     (4, 64, 28, 28, 64, 3, 1, 1),
     (4, 128, 28, 28, 128, 3, 2, 1),
+    (4, 3, 64, 64, 64, 7, 2, 3),     # stem: Cin=3 scalar-gather B path
+    (2, 512, 14, 14, 512, 3, 2, 1),  # deep layer, Cin%64==0 aligned path
+    (4, 256, 14, 14, 1024, 1, 1, 0),
 ])
 def test_mfma_conv_wgrad_vs_miopen(dev, shape):
     torch.manual_seed(5)
