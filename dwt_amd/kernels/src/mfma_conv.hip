@@ -195,6 +195,32 @@ DEV_INLINE void load_a(const bf16* __restrict__ x, const ConvParams& cp,
         rg.c[c] = *reinterpret_cast<const uint4*>(
             x + rc.base[c] + ((int64_t)h * cp.W + w) * cp.Cin + ci);
       }
+    } else if (cp.Cin == 4) {
+      // stem path (3-channel input zero-padded to 4 on the host): a chunk
+      // of 8 halves = two horizontally adjacent pixels; when both are in
+      // bounds on the same kernel row it is ONE 16-B load
+      const int rs0 = kg >> 2;          // kg is 8-aligned -> ci = 0
+      const int rs1 = rs0 + 1;
+      const int s0 = rs0 % cp.KW, r0 = rs0 / cp.KW;
+      const int h0 = rc.h0[c] + r0, w0 = rc.w0[c] + s0;
+      const bool in0 = h0 >= 0 && h0 < cp.H && w0 >= 0 && w0 < cp.W;
+      if (in0 && s0 + 1 < cp.KW && w0 + 1 < cp.W && kg + 4 < cp.K) {
+        rg.c[c] = *reinterpret_cast<const uint4*>(
+            x + rc.base[c] + ((int64_t)h0 * cp.W + w0) * 4);
+      } else {
+        uint2 lo = make_uint2(0, 0), hi = make_uint2(0, 0);
+        if (in0)
+          lo = *reinterpret_cast<const uint2*>(
+              x + rc.base[c] + ((int64_t)h0 * cp.W + w0) * 4);
+        if (kg + 4 < cp.K) {
+          const int s1 = rs1 % cp.KW, r1 = rs1 / cp.KW;
+          const int h1 = rc.h0[c] + r1, w1 = rc.w0[c] + s1;
+          if (h1 >= 0 && h1 < cp.H && w1 >= 0 && w1 < cp.W)
+            hi = *reinterpret_cast<const uint2*>(
+                x + rc.base[c] + ((int64_t)h1 * cp.W + w1) * 4);
+        }
+        rg.c[c] = make_uint4(lo.x, lo.y, hi.x, hi.y);
+      }
     } else {
       unsigned short tmp[8];
 #pragma unroll
@@ -393,8 +419,16 @@ __global__ __launch_bounds__(THREADS, 2) void conv_implicit_gemm_kernel(
 
 
 constexpr int WBM = 64, WBN = 64, WBK = 64;
-constexpr int WPITCH = WBK + 8;  // halves; +16 B keeps the strided
-                                 // transpose writes off one bank
+constexpr int WPITCH = WBK;  // no pad: the chunk XOR swizzle below spreads
+                             // banks instead
+
+// LDS transpose swizzle: element (row, k) lives at
+//   row*WPITCH + ((k>>3) ^ wswz(row))*8 + (k&7)
+// wswz varies with BOTH row%8 and row/8, so the transpose's strided scalar
+// writes (8 rows x 8 k per wave instruction) spread across all banks
+// (un-swizzled they land 16-way conflicted), while b128 fragment reads of
+// 8-aligned k-chunks stay contiguous.
+DEV_INLINE int wswz(int row) { return ((row >> 3) ^ row) & 7; }
 
 template <bool ALIGNED_B>
 __global__ __launch_bounds__(256, 4) void wgrad_splitk_kernel(
@@ -454,7 +488,9 @@ __global__ __launch_bounds__(256, 4) void wgrad_splitk_kernel(
         for (int j = 0; j < 8; ++j) v[j] = 0;
       }
 #pragma unroll
-      for (int j = 0; j < 8; ++j) lds_a[(co + j) * WPITCH + krow] = v[j];
+      for (int j = 0; j < 8; ++j)
+        lds_a[(co + j) * WPITCH + (((krow >> 3) ^ wswz(co + j)) << 3) +
+              (krow & 7)] = v[j];
     }
     // ---- load x-patch tile [WBK k][WBN rsci], transpose into lds_b ----
 #pragma unroll
@@ -496,18 +532,23 @@ __global__ __launch_bounds__(256, 4) void wgrad_splitk_kernel(
         }
       }
 #pragma unroll
-      for (int j = 0; j < 8; ++j) lds_b[(nc + j) * WPITCH + krow] = v[j];
+      for (int j = 0; j < 8; ++j)
+        lds_b[(nc + j) * WPITCH + (((krow >> 3) ^ wswz(nc + j)) << 3) +
+              (krow & 7)] = v[j];
     }
     __syncthreads();
 #pragma unroll
     for (int kstep = 0; kstep < WBK; kstep += 32) {
       short8 afrag[2], bfrag[2];
+      const int kb = (kstep + frag_koff) >> 3;  // 8-aligned chunk index
 #pragma unroll
       for (int i = 0; i < 2; ++i) {
+        const int ra = wm + i * 16 + frag_row;
+        const int rb = wn + i * 16 + frag_row;
         afrag[i] = *reinterpret_cast<const short8*>(
-            lds_a + (wm + i * 16 + frag_row) * WPITCH + kstep + frag_koff);
+            lds_a + ra * WPITCH + ((kb ^ wswz(ra)) << 3));
         bfrag[i] = *reinterpret_cast<const short8*>(
-            lds_b + (wn + i * 16 + frag_row) * WPITCH + kstep + frag_koff);
+            lds_b + rb * WPITCH + ((kb ^ wswz(rb)) << 3));
       }
 #pragma unroll
       for (int i = 0; i < 2; ++i)
@@ -543,6 +584,158 @@ __global__ void f32_to_bf16_kernel(const float* __restrict__ src,
   const int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
   if (i < n)
     *reinterpret_cast<unsigned short*>(dst + i) = f_to_bf16(src[i]);
+}
+
+// ===========================================================================
+// Class-decomposed strided dgrad.  For stride s, dx positions split into s*s
+// residue classes (h%s, w%s); each class has a FIXED subset of (r,s) taps
+// that satisfy the stride-divisibility, so the per-class GEMM runs only the
+// useful K (ntaps*Cdy) instead of wading through KH*KW*Cdy that is ~3/4
+// zeros (the single-kernel gather measured 0.13-0.22x library on the R50
+// stride-2 shapes).  One launch per class; 128x128 tile as the main kernel.
+// ===========================================================================
+
+struct DgradClsParams {
+  int H, W, Cin, P, Q, Cdy, Kfull;  // Kfull = KH*KW*Cdy (wd row pitch)
+  int ch, cw, stride;
+  int Hc, Wc;              // class spatial extent
+  int64_t M;               // N*Hc*Wc
+  int K;                   // ntaps*Cdy
+  int cShift;              // log2(Cdy), Cdy % 8 == 0 guaranteed
+  int dp[16], dq[16], rsIdx[16];
+};
+
+__global__ __launch_bounds__(THREADS, 2) void dgrad_cls_kernel(
+    const bf16* __restrict__ dy, const bf16* __restrict__ wd,
+    bf16* __restrict__ dx, DgradClsParams gp, int mtiles, int ntiles) {
+  __shared__ unsigned short lds_a[BM * LDS_PITCH];
+  __shared__ unsigned short lds_b[BN * LDS_PITCH];
+  const int bid = blockIdx.x;
+  const int64_t m0 = (int64_t)(bid / ntiles) * BM;
+  const int n0 = (bid % ntiles) * BN;
+  const int wave = threadIdx.x / 64;
+  const int lane = threadIdx.x % 64;
+  const int wm = (wave / 2) * 64;
+  const int wn = (wave % 2) * 64;
+
+  floatx4 acc[4][4];
+#pragma unroll
+  for (int i = 0; i < 4; ++i)
+#pragma unroll
+    for (int j = 0; j < 4; ++j) acc[i][j] = (floatx4){0.f, 0.f, 0.f, 0.f};
+
+  // row cache: class position m -> (n, hh, ww)
+  int rcP[4], rcQ[4];
+  int64_t rbase[4];
+  bool rvalid[4];
+  {
+    const int t = threadIdx.x;
+#pragma unroll
+    for (int c = 0; c < 4; ++c) {
+      const int idx = t + c * THREADS;
+      const int row = idx / (BK / 8);
+      const int64_t m = m0 + row;
+      rvalid[c] = m < gp.M;
+      const int64_t mm = rvalid[c] ? m : 0;
+      const int ww = (int)(mm % gp.Wc);
+      const int64_t nh = mm / gp.Wc;
+      const int hh = (int)(nh % gp.Hc);
+      const int n = (int)(nh / gp.Hc);
+      rcP[c] = hh;
+      rcQ[c] = ww;
+      rbase[c] = (int64_t)n * gp.P * gp.Q * gp.Cdy;
+    }
+  }
+
+  const int frag_row = lane % 16;
+  const int frag_koff = (lane / 16) * 8;
+  const int nk = (gp.K + BK - 1) / BK;
+  const int t = threadIdx.x;
+
+  for (int kt = 0; kt < nk; ++kt) {
+    const int k0 = kt * BK;
+    // A: dy gather
+#pragma unroll
+    for (int c = 0; c < 4; ++c) {
+      const int idx = t + c * THREADS;
+      const int kc = (idx % (BK / 8)) * 8;
+      const int kg = k0 + kc;
+      uint4 val = make_uint4(0, 0, 0, 0);
+      if (rvalid[c] && kg < gp.K) {
+        const int tap = kg >> gp.cShift;
+        const int co = kg & (gp.Cdy - 1);
+        const int pp = rcP[c] + gp.dp[tap];
+        const int qq = rcQ[c] + gp.dq[tap];
+        if (pp >= 0 && pp < gp.P && qq >= 0 && qq < gp.Q)
+          val = *reinterpret_cast<const uint4*>(
+              dy + rbase[c] + ((int64_t)pp * gp.Q + qq) * gp.Cdy + co);
+      }
+      const int row = (t + c * THREADS) / (BK / 8);
+      *reinterpret_cast<uint4*>(lds_a + row * LDS_PITCH + kc) = val;
+    }
+    // B: wd rows (ci), k -> (tap, co) -> wd[n][rsIdx[tap]*Cdy + co]
+#pragma unroll
+    for (int c = 0; c < 4; ++c) {
+      const int idx = t + c * THREADS;
+      const int row = idx / (BK / 8);
+      const int kc = (idx % (BK / 8)) * 8;
+      const int kg = k0 + kc;
+      const int n = n0 + row;
+      uint4 val = make_uint4(0, 0, 0, 0);
+      if (n < gp.Cin && kg < gp.K) {
+        const int tap = kg >> gp.cShift;
+        const int co = kg & (gp.Cdy - 1);
+        val = *reinterpret_cast<const uint4*>(
+            wd + (int64_t)n * gp.Kfull + gp.rsIdx[tap] * gp.Cdy + co);
+      }
+      *reinterpret_cast<uint4*>(lds_b + row * LDS_PITCH + kc) = val;
+    }
+    __syncthreads();
+#pragma unroll
+    for (int ks = 0; ks < BK; ks += 32) {
+      short8 afrag[4], bfrag[4];
+#pragma unroll
+      for (int i = 0; i < 4; ++i) {
+        afrag[i] = *reinterpret_cast<const short8*>(
+            lds_a + (wm + i * 16 + frag_row) * LDS_PITCH + ks + frag_koff);
+        bfrag[i] = *reinterpret_cast<const short8*>(
+            lds_b + (wn + i * 16 + frag_row) * LDS_PITCH + ks + frag_koff);
+      }
+#pragma unroll
+      for (int i = 0; i < 4; ++i)
+#pragma unroll
+        for (int j = 0; j < 4; ++j)
+          acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              afrag[i], bfrag[j], acc[i][j], 0, 0, 0);
+    }
+    __syncthreads();
+  }
+
+  // epilogue: scatter to dx[n, hh*s+ch, ww*s+cw, ci]
+  const int erow = (lane / 16) * 4;
+  const int ecol = lane % 16;
+#pragma unroll
+  for (int i = 0; i < 4; ++i) {
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      const int64_t m = m0 + wm + i * 16 + erow + r;
+      if (m >= gp.M) continue;
+      const int ww = (int)(m % gp.Wc);
+      const int64_t nh = m / gp.Wc;
+      const int hh = (int)(nh % gp.Hc);
+      const int n = (int)(nh / gp.Hc);
+      const int h = hh * gp.stride + gp.ch;
+      const int w = ww * gp.stride + gp.cw;
+      const int64_t obase = (((int64_t)n * gp.H + h) * gp.W + w) * gp.Cin;
+#pragma unroll
+      for (int j = 0; j < 4; ++j) {
+        const int col = n0 + wn + j * 16 + ecol;
+        if (col >= gp.Cin) continue;
+        *reinterpret_cast<unsigned short*>(dx + obase + col) =
+            f_to_bf16(acc[i][j][r]);
+      }
+    }
+  }
 }
 
 }  // namespace dwtmm
@@ -616,9 +809,9 @@ void mfma_conv2d_fwd(Tensor x, Tensor wgt, Tensor bias, Tensor out,
   const int ntiles = (Cout + dwtmm::BN - 1) / dwtmm::BN;
   const bool gemm_fast = (KH == 1 && KW == 1 && stride == 1 && pad == 0 &&
                           Cin % 8 == 0);
-  // pipelined (double-buffered, issue-early) staging for every medium/long-K
-  // shape; the gather path's per-k decode now reuses the per-tile row cache
-  const bool pipe = cp.K >= 4 * dwtmm::BK;
+  // pipelined staging pays only on the dense path (measured: the gather
+  // path's extra register pressure under PIPE regressed l3.conv2 0.91->0.60)
+  const bool pipe = gemm_fast && cp.K >= 4 * dwtmm::BK;
   const bool swz = gemm_fast && ntiles >= 8 && mtiles * ntiles >= 16;
   auto run = [&](auto fastc, auto reluc, auto biasc) {
     auto launch = [&](auto pipec, auto swzc) {
@@ -670,7 +863,7 @@ void mfma_conv2d_dgrad(Tensor dy, Tensor wd, Tensor dx, int64_t N, int64_t H,
   // pipelined/swizzled dense path instead of the per-chunk gather decode
   const bool gemm_fast = (KH == 1 && KW == 1 && stride == 1 && pad == 0 &&
                           Cdy % 8 == 0);
-  const bool pipe = cp.K >= 4 * dwtmm::BK;
+  const bool pipe = gemm_fast && cp.K >= 4 * dwtmm::BK;
   const bool swz = gemm_fast && ntiles >= 8 && mtiles * ntiles >= 16;
   auto launch = [&](auto modec, auto pipec, auto swzc) {
     hipLaunchKernelGGL(
@@ -689,6 +882,41 @@ void mfma_conv2d_dgrad(Tensor dy, Tensor wd, Tensor dx, int64_t N, int64_t H,
                 else launch(DEN{}, std::true_type{}, std::false_type{}); }
     else { if (swz) launch(DEN{}, std::false_type{}, std::true_type{});
            else launch(DEN{}, std::false_type{}, std::false_type{}); }
+  } else if (stride > 1 && log2_if_pow2(Cdy) >= 0 && Cdy >= 8) {
+    // class decomposition: one launch per (h%stride, w%stride) residue
+    for (int ch = 0; ch < stride; ++ch) {
+      for (int cw = 0; cw < stride; ++cw) {
+        dwtmm::DgradClsParams gp{};
+        gp.H = H; gp.W = W; gp.Cin = Cin; gp.P = P; gp.Q = Q;
+        gp.Cdy = Cdy; gp.Kfull = KH * KW * Cdy;
+        gp.ch = ch; gp.cw = cw; gp.stride = stride;
+        gp.Hc = (int)((H - ch + stride - 1) / stride);
+        gp.Wc = (int)((W - cw + stride - 1) / stride);
+        gp.M = (int64_t)N * gp.Hc * gp.Wc;
+        gp.cShift = log2_if_pow2(Cdy);
+        int rl[16], sl[16], nr = 0, ns = 0;
+        for (int r = 0; r < KH; ++r)
+          if ((ch + pad - r) % stride == 0 && nr < 4) rl[nr++] = r;
+        for (int s = 0; s < KW; ++s)
+          if ((cw + pad - s) % stride == 0 && ns < 4) sl[ns++] = s;
+        int ntaps = 0;
+        for (int a = 0; a < nr && ntaps < 16; ++a)
+          for (int b = 0; b < ns && ntaps < 16; ++b) {
+            gp.dp[ntaps] = (int)((ch + pad - rl[a]) / stride);
+            gp.dq[ntaps] = (int)((cw + pad - sl[b]) / stride);
+            gp.rsIdx[ntaps] = rl[a] * KW + sl[b];
+            ++ntaps;
+          }
+        gp.K = ntaps * Cdy;
+        if (ntaps == 0 || gp.M == 0) continue;
+        const int cm = (int)((gp.M + dwtmm::BM - 1) / dwtmm::BM);
+        hipLaunchKernelGGL(dwtmm::dgrad_cls_kernel, dim3(cm * ntiles),
+                           dim3(dwtmm::THREADS), 0, dwtmm_stream(),
+                           (const c10::BFloat16*)dy.data_ptr(),
+                           (const c10::BFloat16*)wd.data_ptr(),
+                           (c10::BFloat16*)dx.data_ptr(), gp, cm, ntiles);
+      }
+    }
   } else {
     if (pipe) launch(DGR{}, std::true_type{}, std::false_type{});
     else launch(DGR{}, std::false_type{}, std::false_type{});

@@ -154,7 +154,7 @@ class ResNetDWT(nn.Module):
         self.inplanes = 64
         self.whiten_mode = whiten_mode
 
-        self.conv1 = nn.Conv2d(3, 64, kernel_size=7, stride=2, padding=3, bias=False)
+        self.conv1 = _conv_cls()(3, 64, kernel_size=7, stride=2, padding=3, bias=False)
         (self.bns1, self.bnt1, self.bnt1_aug), self.gamma1, self.beta1 = \
             _make_wh_branches(64, group_size, bn_dict, "bn1", whiten_mode)
         self.maxpool = MaxPool2dDWT(kernel_size=3, stride=2, padding=1)

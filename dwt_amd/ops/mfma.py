@@ -92,6 +92,18 @@ def conv2d_fwd(x: torch.Tensor, weight: torch.Tensor,
     assert x.is_contiguous(memory_format=torch.channels_last) or x.shape[1] == 1
     n, cin, h, w = x.shape
     cout, _, kh, kw = weight.shape
+    if cin == 3:
+        # stem: zero-pad to 4 channels so the kernel's paired-pixel 16-B
+        # path applies (NHWC C=4 chunks); weight padded to match
+        x4 = torch.empty(n, 4, h, w, device=x.device, dtype=x.dtype,
+                         memory_format=torch.channels_last)
+        x4[:, 3] = 0
+        x4[:, :3] = x
+        w4 = torch.empty(cout, 4, kh, kw, device=x.device, dtype=weight.dtype,
+                         memory_format=torch.channels_last)
+        w4[:, 3] = 0
+        w4[:, :3] = weight.detach()
+        x, weight, cin = x4, w4, 4
     p = (h + 2 * padding - kh) // stride + 1
     q = (w + 2 * padding - kw) // stride + 1
     out = torch.empty(n, cout, p, q, device=x.device, dtype=torch.bfloat16,
@@ -166,7 +178,8 @@ class MFMAConv2d(nn.Conv2d):
     DWT_AMD_CONV=hip."""
 
     def forward(self, x):
-        if x.is_cuda and x.dtype == torch.bfloat16 and x.shape[1] >= 8 \
+        if x.is_cuda and x.dtype == torch.bfloat16 \
+                and (x.shape[1] >= 8 or x.shape[1] == 3) \
                 and x.is_contiguous(memory_format=torch.channels_last) \
                 and self.stride[0] == self.stride[1] \
                 and self.padding[0] == self.padding[1]:
