@@ -51,6 +51,11 @@ def main():
     ap = argparse.ArgumentParser()
     ap.add_argument("--out", default="profiles/conv_bench.md")
     ap.add_argument("--iters", type=int, default=20)
+    ap.add_argument("--batch", type=int, default=192,
+                    help="N (the flagship bench step runs N=1536 = 3x512)")
+    ap.add_argument("--only", default="",
+                    help="comma-separated substring filter on shape names")
+    ap.add_argument("--passes", default="fwd,dgrad,wgrad")
     args = ap.parse_args()
     assert torch.cuda.is_available()
     from dwt_amd.ops.mfma import conv2d_fwd, mfma_gemm
@@ -72,7 +77,12 @@ def main():
             g, x, wt, None, [stride, stride], [pad, pad], [1, 1], False,
             [0, 0], 1, [False, True, False])[1]
 
+    only = [t for t in args.only.split(",") if t]
+    passes_on = args.passes.split(",")
     for name, n, cin, h, w, cout, k, stride, pad in R50_SHAPES:
+        n = args.batch
+        if only and not any(t in name for t in only):
+            continue
         x = torch.randn(n, cin, h, w, device=dev).to(torch.bfloat16) \
             .contiguous(memory_format=torch.channels_last)
         wt = (torch.randn(cout, cin, k, k, device=dev) * 0.05).to(torch.bfloat16) \
@@ -92,6 +102,8 @@ def main():
              lambda: lib_wgrad(g, x, wt, stride, pad)),
         ]
         for pname, ours_fn, lib_fn in passes:
+            if pname not in passes_on:
+                continue
             if pname == "dgrad" and name == "stem7x7":
                 continue  # stem never needs dx
             t_ours = timeit(ours_fn, args.iters)
