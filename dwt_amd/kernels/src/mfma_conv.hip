@@ -470,38 +470,45 @@ __global__ __launch_bounds__(256, 4) void wgrad_splitk_kernel(
   const int frag_row = lane % 16;
   const int frag_koff = (lane / 16) * 8;
 
+  // thread t owns channel chunk (t%8)*8 and the k-pair 2*(t/8): one pass of
+  // 256 threads covers a full 64x64 operand tile, and adjacent k values
+  // pack into ds_write_b32 (the per-element u16 transpose stores measured
+  // 4x the LDS write cost and capped the kernel at ~180 TF)
+  const int cch = (t % 8) * 8;
+  const int kp = 2 * (t / 8);
+
   for (int64_t k0 = ks; k0 < ke; k0 += WBK) {
-    // ---- load dy tile [WBK k][WBM co] coalesced, transpose into lds_a ----
-    // thread t covers (krow = t/8, co chunk = (t%8)*8), two passes of 32 rows
-#pragma unroll
-    for (int pass = 0; pass < 2; ++pass) {
-      const int krow = (t / 8) + pass * 32;
-      const int co = (t % 8) * 8;
-      const int64_t k = k0 + krow;
-      unsigned short v[8];
-      if (k < ke && m0 + co < cp.N) {  // cp.N = Cout here
-        const uint4 u = *reinterpret_cast<const uint4*>(
-            dy + k * cp.N + m0 + co);
-        *reinterpret_cast<uint4*>(v) = u;
-      } else {
-#pragma unroll
-        for (int j = 0; j < 8; ++j) v[j] = 0;
+    // ---- dy tile [WBK k][WBM co] -> lds_a[co][k] ----
+    {
+      const int64_t ka = k0 + kp;
+      uint4 u0 = make_uint4(0, 0, 0, 0), u1 = make_uint4(0, 0, 0, 0);
+      if (m0 + cch < cp.N) {  // cp.N = Cout here
+        if (ka < ke)
+          u0 = *reinterpret_cast<const uint4*>(dy + ka * cp.N + m0 + cch);
+        if (ka + 1 < ke)
+          u1 = *reinterpret_cast<const uint4*>(dy + (ka + 1) * cp.N + m0 + cch);
       }
+      const unsigned short* v0 = reinterpret_cast<const unsigned short*>(&u0);
+      const unsigned short* v1 = reinterpret_cast<const unsigned short*>(&u1);
 #pragma unroll
-      for (int j = 0; j < 8; ++j)
-        lds_a[(co + j) * WPITCH + (((krow >> 3) ^ wswz(co + j)) << 3) +
-              (krow & 7)] = v[j];
+      for (int j = 0; j < 8; ++j) {
+        const int row = cch + j;
+        const unsigned int pk = (unsigned int)v0[j] | ((unsigned int)v1[j] << 16);
+        *reinterpret_cast<unsigned int*>(
+            lds_a + row * WPITCH + (((kp >> 3) ^ wswz(row)) << 3) + (kp & 7)) = pk;
+      }
     }
-    // ---- load x-patch tile [WBK k][WBN rsci], transpose into lds_b ----
+    // ---- x-patch tile [WBK k][WBN rsci] -> lds_b[rsci][k] ----
+    {
+      const int64_t kb0 = k0 + kp;
+      unsigned short v0[8], v1[8];
 #pragma unroll
-    for (int pass = 0; pass < 2; ++pass) {
-      const int krow = (t / 8) + pass * 32;
-      const int nc = (t % 8) * 8;       // rsci offset within tile
-      const int64_t k = k0 + krow;
-      unsigned short v[8];
+      for (int j = 0; j < 8; ++j) { v0[j] = 0; v1[j] = 0; }
 #pragma unroll
-      for (int j = 0; j < 8; ++j) v[j] = 0;
-      if (k < ke) {
+      for (int half = 0; half < 2; ++half) {
+        const int64_t k = kb0 + half;
+        unsigned short* v = half ? v1 : v0;
+        if (k >= ke) continue;
         const int q = (int)(k % cp.Q);
         const int64_t np = k / cp.Q;
         const int p = (int)(np % cp.P);
@@ -509,15 +516,13 @@ __global__ __launch_bounds__(256, 4) void wgrad_splitk_kernel(
         if (ALIGNED_B) {
           const int h = p * cp.stride - cp.pad + rB;
           const int w = q * cp.stride - cp.pad + sB;
-          if (h >= 0 && h < cp.H && w >= 0 && w < cp.W) {
-            const uint4 u = *reinterpret_cast<const uint4*>(
-                x + (((int64_t)n * cp.H + h) * cp.W + w) * cp.Cin + ciB + nc);
-            *reinterpret_cast<uint4*>(v) = u;
-          }
+          if (h >= 0 && h < cp.H && w >= 0 && w < cp.W)
+            *reinterpret_cast<uint4*>(v) = *reinterpret_cast<const uint4*>(
+                x + (((int64_t)n * cp.H + h) * cp.W + w) * cp.Cin + ciB + cch);
         } else {
 #pragma unroll
           for (int j = 0; j < 8; ++j) {
-            const int nr = n0 + nc + j;
+            const int nr = n0 + cch + j;
             if (nr >= cp.Cout) break;   // cp.Cout = KH*KW*Cin columns
             const int ci = nr % cp.Cin;
             const int rs = nr / cp.Cin;
@@ -532,9 +537,12 @@ __global__ __launch_bounds__(256, 4) void wgrad_splitk_kernel(
         }
       }
 #pragma unroll
-      for (int j = 0; j < 8; ++j)
-        lds_b[(nc + j) * WPITCH + (((krow >> 3) ^ wswz(nc + j)) << 3) +
-              (krow & 7)] = v[j];
+      for (int j = 0; j < 8; ++j) {
+        const int row = cch + j;
+        const unsigned int pk = (unsigned int)v0[j] | ((unsigned int)v1[j] << 16);
+        *reinterpret_cast<unsigned int*>(
+            lds_b + row * WPITCH + (((kp >> 3) ^ wswz(row)) << 3) + (kp & 7)) = pk;
+      }
     }
     __syncthreads();
 #pragma unroll
