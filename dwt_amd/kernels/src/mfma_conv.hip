@@ -418,20 +418,20 @@ __global__ __launch_bounds__(THREADS, 2) void conv_implicit_gemm_kernel(
 
 
 
-constexpr int WBM = 64, WBN = 64, WBK = 64;
+constexpr int WBM = 128, WBN = 128, WBK = 64;
 constexpr int WPITCH = WBK;  // no pad: the chunk XOR swizzle below spreads
                              // banks instead
 
 // LDS transpose swizzle: element (row, k) lives at
 //   row*WPITCH + ((k>>3) ^ wswz(row))*8 + (k&7)
-// wswz varies with BOTH row%8 and row/8, so the transpose's strided scalar
-// writes (8 rows x 8 k per wave instruction) spread across all banks
+// wswz varies with BOTH row%8 and row/8, so the transpose's strided stores
+// (8 rows x one k-pair per wave instruction) spread across all banks
 // (un-swizzled they land 16-way conflicted), while b128 fragment reads of
 // 8-aligned k-chunks stay contiguous.
 DEV_INLINE int wswz(int row) { return ((row >> 3) ^ row) & 7; }
 
 template <bool ALIGNED_B>
-__global__ __launch_bounds__(256, 4) void wgrad_splitk_kernel(
+__global__ __launch_bounds__(256, 3) void wgrad_splitk_kernel(
     const bf16* __restrict__ dy, const bf16* __restrict__ x,
     float* __restrict__ ws, ConvParams cp, int mtiles, int ntiles,
     int splitk, int64_t klen) {
@@ -449,64 +449,65 @@ __global__ __launch_bounds__(256, 4) void wgrad_splitk_kernel(
   const int t = threadIdx.x;
   const int wave = t / 64;
   const int lane = t % 64;
-  const int wm = (wave / 2) * 32;
-  const int wn = (wave % 2) * 32;
+  const int wm = (wave / 2) * 64;
+  const int wn = (wave % 2) * 64;
 
-  // B-tile tap decode (single (r,s) per tile when ALIGNED_B)
-  int rB = 0, sB = 0, ciB = 0;
-  if (ALIGNED_B) {
-    const int rs = n0 / cp.Cin;
-    sB = rs % cp.KW;
-    rB = rs / cp.KW;
-    ciB = n0 % cp.Cin;
-  }
-
-  floatx4 acc[2][2];
+  floatx4 acc[4][4];
 #pragma unroll
-  for (int i = 0; i < 2; ++i)
+  for (int i = 0; i < 4; ++i)
 #pragma unroll
-    for (int j = 0; j < 2; ++j) acc[i][j] = (floatx4){0.f, 0.f, 0.f, 0.f};
+    for (int j = 0; j < 4; ++j) acc[i][j] = (floatx4){0.f, 0.f, 0.f, 0.f};
 
   const int frag_row = lane % 16;
   const int frag_koff = (lane / 16) * 8;
 
-  // thread t owns channel chunk (t%8)*8 and the k-pair 2*(t/8): one pass of
-  // 256 threads covers a full 64x64 operand tile, and adjacent k values
-  // pack into ds_write_b32 (the per-element u16 transpose stores measured
-  // 4x the LDS write cost and capped the kernel at ~180 TF)
+  // thread t owns channel chunk (t%8)*8 (two 64-row passes) and the k-pair
+  // 2*(t/8): adjacent k values pack into ds_write_b32 (per-element u16
+  // transpose stores cost 4x the LDS write cycles)
   const int cch = (t % 8) * 8;
   const int kp = 2 * (t / 8);
 
   for (int64_t k0 = ks; k0 < ke; k0 += WBK) {
+    const int64_t ka = k0 + kp;
     // ---- dy tile [WBK k][WBM co] -> lds_a[co][k] ----
-    {
-      const int64_t ka = k0 + kp;
+#pragma unroll
+    for (int pass = 0; pass < 2; ++pass) {
+      const int c2 = cch + pass * 64;
       uint4 u0 = make_uint4(0, 0, 0, 0), u1 = make_uint4(0, 0, 0, 0);
-      if (m0 + cch < cp.N) {  // cp.N = Cout here
+      if (m0 + c2 < cp.N) {  // cp.N = Cout here
         if (ka < ke)
-          u0 = *reinterpret_cast<const uint4*>(dy + ka * cp.N + m0 + cch);
+          u0 = *reinterpret_cast<const uint4*>(dy + ka * cp.N + m0 + c2);
         if (ka + 1 < ke)
-          u1 = *reinterpret_cast<const uint4*>(dy + (ka + 1) * cp.N + m0 + cch);
+          u1 = *reinterpret_cast<const uint4*>(dy + (ka + 1) * cp.N + m0 + c2);
       }
       const unsigned short* v0 = reinterpret_cast<const unsigned short*>(&u0);
       const unsigned short* v1 = reinterpret_cast<const unsigned short*>(&u1);
 #pragma unroll
       for (int j = 0; j < 8; ++j) {
-        const int row = cch + j;
+        const int row = c2 + j;
         const unsigned int pk = (unsigned int)v0[j] | ((unsigned int)v1[j] << 16);
         *reinterpret_cast<unsigned int*>(
             lds_a + row * WPITCH + (((kp >> 3) ^ wswz(row)) << 3) + (kp & 7)) = pk;
       }
     }
     // ---- x-patch tile [WBK k][WBN rsci] -> lds_b[rsci][k] ----
-    {
-      const int64_t kb0 = k0 + kp;
+#pragma unroll
+    for (int pass = 0; pass < 2; ++pass) {
+      const int c2 = cch + pass * 64;
       unsigned short v0[8], v1[8];
 #pragma unroll
       for (int j = 0; j < 8; ++j) { v0[j] = 0; v1[j] = 0; }
+      // when Cin % 64 == 0 the 8-row chunk (64-aligned) sits in ONE tap
+      int rB = 0, sB = 0, ciB = 0;
+      if (ALIGNED_B) {
+        const int rs = (n0 + c2) / cp.Cin;
+        sB = rs % cp.KW;
+        rB = rs / cp.KW;
+        ciB = (n0 + c2) % cp.Cin;
+      }
 #pragma unroll
       for (int half = 0; half < 2; ++half) {
-        const int64_t k = kb0 + half;
+        const int64_t k = ka + half;
         unsigned short* v = half ? v1 : v0;
         if (k >= ke) continue;
         const int q = (int)(k % cp.Q);
@@ -518,11 +519,11 @@ __global__ __launch_bounds__(256, 4) void wgrad_splitk_kernel(
           const int w = q * cp.stride - cp.pad + sB;
           if (h >= 0 && h < cp.H && w >= 0 && w < cp.W)
             *reinterpret_cast<uint4*>(v) = *reinterpret_cast<const uint4*>(
-                x + (((int64_t)n * cp.H + h) * cp.W + w) * cp.Cin + ciB + cch);
+                x + (((int64_t)n * cp.H + h) * cp.W + w) * cp.Cin + ciB);
         } else {
 #pragma unroll
           for (int j = 0; j < 8; ++j) {
-            const int nr = n0 + cch + j;
+            const int nr = n0 + c2 + j;
             if (nr >= cp.Cout) break;   // cp.Cout = KH*KW*Cin columns
             const int ci = nr % cp.Cin;
             const int rs = nr / cp.Cin;
@@ -538,7 +539,7 @@ __global__ __launch_bounds__(256, 4) void wgrad_splitk_kernel(
       }
 #pragma unroll
       for (int j = 0; j < 8; ++j) {
-        const int row = cch + j;
+        const int row = c2 + j;
         const unsigned int pk = (unsigned int)v0[j] | ((unsigned int)v1[j] << 16);
         *reinterpret_cast<unsigned int*>(
             lds_b + row * WPITCH + (((kp >> 3) ^ wswz(row)) << 3) + (kp & 7)) = pk;
@@ -547,10 +548,10 @@ __global__ __launch_bounds__(256, 4) void wgrad_splitk_kernel(
     __syncthreads();
 #pragma unroll
     for (int kstep = 0; kstep < WBK; kstep += 32) {
-      short8 afrag[2], bfrag[2];
+      short8 afrag[4], bfrag[4];
       const int kb = (kstep + frag_koff) >> 3;  // 8-aligned chunk index
 #pragma unroll
-      for (int i = 0; i < 2; ++i) {
+      for (int i = 0; i < 4; ++i) {
         const int ra = wm + i * 16 + frag_row;
         const int rb = wn + i * 16 + frag_row;
         afrag[i] = *reinterpret_cast<const short8*>(
@@ -559,9 +560,9 @@ __global__ __launch_bounds__(256, 4) void wgrad_splitk_kernel(
             lds_b + rb * WPITCH + ((kb ^ wswz(rb)) << 3));
       }
 #pragma unroll
-      for (int i = 0; i < 2; ++i)
+      for (int i = 0; i < 4; ++i)
 #pragma unroll
-        for (int j = 0; j < 2; ++j)
+        for (int j = 0; j < 4; ++j)
           acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
               afrag[i], bfrag[j], acc[i][j], 0, 0, 0);
     }
@@ -572,13 +573,13 @@ __global__ __launch_bounds__(256, 4) void wgrad_splitk_kernel(
   const int erow = (lane / 16) * 4;
   const int ecol = lane % 16;
 #pragma unroll
-  for (int i = 0; i < 2; ++i) {
+  for (int i = 0; i < 4; ++i) {
 #pragma unroll
     for (int r = 0; r < 4; ++r) {
       const int m = m0 + wm + i * 16 + erow + r;
       if (m >= cp.N) continue;
 #pragma unroll
-      for (int j = 0; j < 2; ++j) {
+      for (int j = 0; j < 4; ++j) {
         const int n = n0 + wn + j * 16 + ecol;
         if (n >= cp.Cout) continue;
         atomicAdd(&ws[(int64_t)m * cp.Cout + n], acc[i][j][r]);
