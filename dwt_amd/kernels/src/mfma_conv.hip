@@ -418,7 +418,7 @@ __global__ __launch_bounds__(THREADS, 2) void conv_implicit_gemm_kernel(
 
 
 
-constexpr int WBM = 128, WBN = 128, WBK = 64;
+constexpr int WBK = 64;
 constexpr int WPITCH = WBK;  // no pad: the chunk XOR swizzle below spreads
                              // banks instead
 
@@ -430,33 +430,38 @@ constexpr int WPITCH = WBK;  // no pad: the chunk XOR swizzle below spreads
 // 8-aligned k-chunks stay contiguous.
 DEV_INLINE int wswz(int row) { return ((row >> 3) ^ row) & 7; }
 
-template <bool ALIGNED_B>
-__global__ __launch_bounds__(256, 3) void wgrad_splitk_kernel(
+// WT = square tile side (64 for small Cout/RSC shapes, 128 otherwise —
+// the 64 tile re-reads operands across tiles and goes HBM-bound on big
+// shapes; the 128 tile wastes MFMA work when M or N < 128).
+template <bool ALIGNED_B, int WT>
+__global__ __launch_bounds__(256, WT == 64 ? 4 : 3) void wgrad_splitk_kernel(
     const bf16* __restrict__ dy, const bf16* __restrict__ x,
     float* __restrict__ ws, ConvParams cp, int mtiles, int ntiles,
     int splitk, int64_t klen) {
-  __shared__ unsigned short lds_a[WBM * WPITCH];  // [co][k]
-  __shared__ unsigned short lds_b[WBN * WPITCH];  // [rsci][k]
+  constexpr int NP = WT / 64;   // load passes per operand
+  constexpr int NF = WT / 32;   // MFMA fragments per wave dim
+  __shared__ unsigned short lds_a[WT * WPITCH];  // [co][k]
+  __shared__ unsigned short lds_b[WT * WPITCH];  // [rsci][k]
 
   const int bid = blockIdx.x;
   const int tile = bid / splitk;
   const int kchunk = bid % splitk;
-  const int m0 = (tile / ntiles) * WBM;   // co origin
-  const int n0 = (tile % ntiles) * WBN;   // rsci origin
+  const int m0 = (tile / ntiles) * WT;   // co origin
+  const int n0 = (tile % ntiles) * WT;   // rsci origin
   const int64_t ks = (int64_t)kchunk * klen;
   const int64_t ke = (ks + klen < cp.M) ? ks + klen : cp.M;  // cp.M = real NPQ
 
   const int t = threadIdx.x;
   const int wave = t / 64;
   const int lane = t % 64;
-  const int wm = (wave / 2) * 64;
-  const int wn = (wave % 2) * 64;
+  const int wm = (wave / 2) * (WT / 2);
+  const int wn = (wave % 2) * (WT / 2);
 
-  floatx4 acc[4][4];
+  floatx4 acc[NF][NF];
 #pragma unroll
-  for (int i = 0; i < 4; ++i)
+  for (int i = 0; i < NF; ++i)
 #pragma unroll
-    for (int j = 0; j < 4; ++j) acc[i][j] = (floatx4){0.f, 0.f, 0.f, 0.f};
+    for (int j = 0; j < NF; ++j) acc[i][j] = (floatx4){0.f, 0.f, 0.f, 0.f};
 
   const int frag_row = lane % 16;
   const int frag_koff = (lane / 16) * 8;
@@ -469,9 +474,9 @@ __global__ __launch_bounds__(256, 3) void wgrad_splitk_kernel(
 
   for (int64_t k0 = ks; k0 < ke; k0 += WBK) {
     const int64_t ka = k0 + kp;
-    // ---- dy tile [WBK k][WBM co] -> lds_a[co][k] ----
+    // ---- dy tile [WBK k][WT co] -> lds_a[co][k] ----
 #pragma unroll
-    for (int pass = 0; pass < 2; ++pass) {
+    for (int pass = 0; pass < NP; ++pass) {
       const int c2 = cch + pass * 64;
       uint4 u0 = make_uint4(0, 0, 0, 0), u1 = make_uint4(0, 0, 0, 0);
       if (m0 + c2 < cp.N) {  // cp.N = Cout here
@@ -490,9 +495,9 @@ __global__ __launch_bounds__(256, 3) void wgrad_splitk_kernel(
             lds_a + row * WPITCH + (((kp >> 3) ^ wswz(row)) << 3) + (kp & 7)) = pk;
       }
     }
-    // ---- x-patch tile [WBK k][WBN rsci] -> lds_b[rsci][k] ----
+    // ---- x-patch tile [WBK k][WT rsci] -> lds_b[rsci][k] ----
 #pragma unroll
-    for (int pass = 0; pass < 2; ++pass) {
+    for (int pass = 0; pass < NP; ++pass) {
       const int c2 = cch + pass * 64;
       unsigned short v0[8], v1[8];
 #pragma unroll
@@ -548,10 +553,10 @@ __global__ __launch_bounds__(256, 3) void wgrad_splitk_kernel(
     __syncthreads();
 #pragma unroll
     for (int kstep = 0; kstep < WBK; kstep += 32) {
-      short8 afrag[4], bfrag[4];
+      short8 afrag[NF], bfrag[NF];
       const int kb = (kstep + frag_koff) >> 3;  // 8-aligned chunk index
 #pragma unroll
-      for (int i = 0; i < 4; ++i) {
+      for (int i = 0; i < NF; ++i) {
         const int ra = wm + i * 16 + frag_row;
         const int rb = wn + i * 16 + frag_row;
         afrag[i] = *reinterpret_cast<const short8*>(
@@ -560,9 +565,9 @@ __global__ __launch_bounds__(256, 3) void wgrad_splitk_kernel(
             lds_b + rb * WPITCH + ((kb ^ wswz(rb)) << 3));
       }
 #pragma unroll
-      for (int i = 0; i < 4; ++i)
+      for (int i = 0; i < NF; ++i)
 #pragma unroll
-        for (int j = 0; j < 4; ++j)
+        for (int j = 0; j < NF; ++j)
           acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
               afrag[i], bfrag[j], acc[i][j], 0, 0, 0);
     }
@@ -573,13 +578,13 @@ __global__ __launch_bounds__(256, 3) void wgrad_splitk_kernel(
   const int erow = (lane / 16) * 4;
   const int ecol = lane % 16;
 #pragma unroll
-  for (int i = 0; i < 4; ++i) {
+  for (int i = 0; i < NF; ++i) {
 #pragma unroll
     for (int r = 0; r < 4; ++r) {
       const int m = m0 + wm + i * 16 + erow + r;
       if (m >= cp.N) continue;
 #pragma unroll
-      for (int j = 0; j < 4; ++j) {
+      for (int j = 0; j < NF; ++j) {
         const int n = n0 + wn + j * 16 + ecol;
         if (n >= cp.Cout) continue;
         atomicAdd(&ws[(int64_t)m * cp.Cout + n], acc[i][j][r]);
@@ -948,8 +953,11 @@ void mfma_conv2d_wgrad(Tensor dy, Tensor x, Tensor dw, Tensor ws, int64_t N,
   cp.KH = KH; cp.KW = KW; cp.stride = stride; cp.pad = pad;
   cp.M = (int64_t)N * P * Q;
   cp.Cout = KH * KW * Cin;
-  const int mtiles = (Cout + dwtmm::WBM - 1) / dwtmm::WBM;
-  const int ntiles = (cp.Cout + dwtmm::WBN - 1) / dwtmm::WBN;
+  // 128 tile when both dims fill it (4x less cross-tile operand traffic);
+  // 64 tile otherwise (stem/l1 shapes would waste half the MFMA work)
+  const int WT = (Cout >= 128 && cp.Cout >= 128) ? 128 : 64;
+  const int mtiles = (int)((Cout + WT - 1) / WT);
+  const int ntiles = (cp.Cout + WT - 1) / WT;
   // split K so the grid lands at >=2048 workgroups (256 CUs, several waves
   // deep), k-slabs rounded to whole BK tiles
   int64_t splitk = std::max<int64_t>(1, 2048 / (mtiles * ntiles));
@@ -957,15 +965,19 @@ void mfma_conv2d_wgrad(Tensor dy, Tensor x, Tensor dw, Tensor ws, int64_t N,
   splitk = std::min<int64_t>(splitk, kt);
   const int64_t klen = ((kt + splitk - 1) / splitk) * dwtmm::WBK;
   const bool aligned = (Cin % 64 == 0);
-  auto launch = [&](auto ac) {
+  auto launch = [&](auto ac, auto wtc) {
     hipLaunchKernelGGL(
-        (dwtmm::wgrad_splitk_kernel<decltype(ac)::value>),
+        (dwtmm::wgrad_splitk_kernel<decltype(ac)::value, decltype(wtc)::value>),
         dim3(mtiles * ntiles * splitk), dim3(256), 0, dwtmm_stream(),
         (const c10::BFloat16*)dy.data_ptr(), (const c10::BFloat16*)x.data_ptr(),
         ws.data_ptr<float>(), cp, mtiles, ntiles, (int)splitk, klen);
   };
-  if (aligned) launch(std::true_type{});
-  else launch(std::false_type{});
+  using T64 = std::integral_constant<int, 64>;
+  using T128 = std::integral_constant<int, 128>;
+  if (aligned) { if (WT == 128) launch(std::true_type{}, T128{});
+                 else launch(std::true_type{}, T64{}); }
+  else { if (WT == 128) launch(std::false_type{}, T128{});
+         else launch(std::false_type{}, T64{}); }
   const int64_t n = (int64_t)Cout * cp.Cout;
   hipLaunchKernelGGL(dwtmm::f32_to_bf16_kernel, dim3((n + 255) / 256),
                      dim3(256), 0, dwtmm_stream(), ws.data_ptr<float>(),
