@@ -13,13 +13,15 @@ import torch
 
 from . import dispatch
 
-_FAST_G = (2, 4, 8)
+_FAST_G = (2, 4, 8)      # register-blocked kernels (NCHW + NHWC)
+_GEN_G_MAX = 32          # LDS-tiled generic kernels, NCHW (g=16/32 digits)
 _warned = set()
 
 
 def _whiten_layout(x, c, g):
     """'cl' when x is channels_last and the NHWC kernels support (C, g):
-    the lane map needs GW = min(C,256)/g to divide the 256-thread block."""
+    the lane map needs GW = min(C,256)/g to divide the 256-thread block.
+    g > 8 always routes NCHW (the generic LDS-tiled kernels)."""
     if x.dim() == 4 and x.is_contiguous(memory_format=torch.channels_last) \
             and not x.is_contiguous():
         if c % 64 == 0 and (c <= 256 or c % 256 == 0) and g in (2, 4):
@@ -54,6 +56,20 @@ def _warn_once(key, msg):
     if key not in _warned:
         _warned.add(key)
         warnings.warn(msg)
+
+
+def _sync_world():
+    """World size for cross-rank ('sync') batch statistics."""
+    import torch.distributed as dist
+    if dist.is_available() and dist.is_initialized():
+        return dist, dist.get_world_size()
+    return None, 1
+
+
+def _allreduce(t):
+    import torch.distributed as dist
+    dist.all_reduce(t)
+    return t
 
 
 def _ext():
@@ -99,12 +115,31 @@ class _HipWhitenMulti(torch.autograd.Function):
         gflat = gamma.detach().reshape(c).contiguous() if has_affine else torch.empty(0, device=dev, dtype=x.dtype)
         bflat = beta.detach().reshape(c).contiguous() if has_affine else torch.empty(0, device=dev, dtype=x.dtype)
 
+        stats_sync = cfg.get("stats_sync", False) and use_batch
+        world = 1
+        if stats_sync:
+            _, world = _sync_world()
+        ctx_count = m_count * world  # global positions per branch
+
         out = torch.empty_like(x)
         if use_batch:
             acc = torch.zeros(ng_all * (g + g * g), device=dev, dtype=torch.float32)
             mean = torch.empty(parts * c, device=dev, dtype=torch.float32)
             cov = torch.empty(ng_all, g, g, device=dev, dtype=torch.float32)
-            if layout == "cl":
+            if stats_sync:
+                # split pass: local raw sums -> all-reduce -> finalize over
+                # the GLOBAL batch (SyncBN-style cross-rank statistics)
+                if layout == "cl":
+                    ext.whiten_stats_partial_cl(x, acc, g, c, m_count, parts)
+                else:
+                    for p in range(parts):
+                        ext.whiten_stats_partial(
+                            x[p * b:(p + 1) * b],
+                            acc[p * n_groups * (g + g * g):(p + 1) * n_groups * (g + g * g)], g)
+                if world > 1:
+                    _allreduce(acc)
+                ext.whiten_stats_final(acc, mean, cov, g, ng_all, ctx_count)
+            elif layout == "cl":
                 ext.whiten_stats_cl(x, acc, mean, cov, g, c, m_count, parts)
             else:
                 for p in range(parts):
@@ -158,7 +193,8 @@ class _HipWhitenMulti(torch.autograd.Function):
         ctx.cfg = cfg
         ctx.g = g
         ctx.layout = layout
-        ctx.m_count = m_count
+        ctx.m_count = ctx_count
+        ctx.sync_world = world
         ctx.mean = mean
         ctx.wmat = wmat
         ctx.saved_mat = saved_mat
@@ -212,15 +248,22 @@ class _HipWhitenMulti(torch.autograd.Function):
             gdb = (ctx.gflat.float().unsqueeze(0) * dgb[:, 1]) if ctx.has_affine \
                 else dgb[:, 1]
             gdb = gdb.reshape(-1).contiguous()
+            dW_in = dW
+            if getattr(ctx, "sync_world", 1) > 1:
+                # global-batch statistics: the matrix-function backward sees
+                # the cross-rank sums (dgamma/dbeta stay LOCAL — the DP
+                # gradient all-reduce handles parameter grads)
+                dW_in = _allreduce(dW.clone())
+                gdb = _allreduce(gdb.clone())
             S = torch.empty(ng_all, g, g, device=dev, dtype=torch.float32)
             corr = torch.empty(parts * c, device=dev, dtype=torch.float32)
             inv_m = 1.0 / m_count
             if mode == "chol":
-                ext.matfn_chol_bwd(dW, wmat, ctx.saved_mat, gdb, S, corr,
+                ext.matfn_chol_bwd(dW_in, wmat, ctx.saved_mat, gdb, S, corr,
                                    eps, inv_m)
             else:
                 ys, zs, svals = ctx.saved_mat
-                ext.matfn_ns_bwd(dW, wmat, ys, zs, svals, gdb, S, corr,
+                ext.matfn_ns_bwd(dW_in, wmat, ys, zs, svals, gdb, S, corr,
                                  eps, inv_m, ns_iters)
         else:
             S = torch.empty(0, device=dev)
@@ -251,10 +294,10 @@ class _HipWhitenMulti(torch.autograd.Function):
 
 def whiten_multi(x, gamma, beta, running_means, running_vars, cfg):
     g = x.shape[1] // cfg["num_groups"]
-    if g not in _FAST_G or x.dtype not in (torch.float32, torch.bfloat16):
+    if g > _GEN_G_MAX or x.dtype not in (torch.float32, torch.bfloat16):
         _warn_once(("wh", g, x.dtype),
                    f"dwt_amd: whiten_multi g={g} dtype={x.dtype} uses the "
-                   "torch path (HIP fast path covers g in {2,4,8}, fp32/bf16)")
+                   "torch path (HIP kernels cover g <= 32, fp32/bf16)")
         from ..ops.functional import WhitenMulti
         return WhitenMulti.apply(x, gamma, beta, running_means, running_vars, cfg)
     return _HipWhitenMulti.apply(x, gamma, beta, running_means, running_vars, cfg)
@@ -285,13 +328,30 @@ class _HipBatchNormMulti(torch.autograd.Function):
         gflat = gamma.detach().reshape(c).contiguous() if has_affine else torch.empty(0, device=dev, dtype=x.dtype)
         bflat = beta.detach().reshape(c).contiguous() if has_affine else torch.empty(0, device=dev, dtype=x.dtype)
 
+        stats_sync = cfg.get("stats_sync", False) and use_batch
+        world = 1
+        if stats_sync:
+            _, world = _sync_world()
+        cnt_total = cnt * world
+
         out = torch.empty_like(x)
         if use_batch:
             acc = torch.zeros(parts * 2 * c, device=dev, dtype=torch.float32)
             mean = torch.empty(parts * c, device=dev, dtype=torch.float32)
             istd = torch.empty(parts * c, device=dev, dtype=torch.float32)
             var_unb = torch.empty(parts * c, device=dev, dtype=torch.float32)
-            if layout == "cl":
+            if stats_sync:
+                if layout == "cl":
+                    ext.bn_stats_partial_cl(x, acc, c, cnt, parts)
+                else:
+                    for p in range(parts):
+                        ext.bn_stats_partial(x[p * b:(p + 1) * b],
+                                             acc[p * 2 * c:(p + 1) * 2 * c])
+                if world > 1:
+                    _allreduce(acc)
+                ext.bn_stats_final(acc, mean, istd, var_unb, c, parts,
+                                   cnt_total, eps)
+            elif layout == "cl":
                 ext.bn_stats_cl(x, acc, mean, istd, var_unb, c, cnt, eps, parts)
             else:
                 for p in range(parts):
@@ -331,6 +391,8 @@ class _HipBatchNormMulti(torch.autograd.Function):
         ctx.cfg = cfg
         ctx.layout = layout
         ctx.cnt = cnt
+        ctx.cnt_total = cnt_total
+        ctx.sync_world = world
         ctx.spatial = spatial
         ctx.mean = mean
         ctx.istd = istd
@@ -358,12 +420,17 @@ class _HipBatchNormMulti(torch.autograd.Function):
 
         sums = torch.zeros(parts, 2, c, device=dev, dtype=torch.float32)
         dx = torch.empty_like(x)
+        cnt_total = getattr(ctx, "cnt_total", ctx.cnt)
         if layout == "cl":
             ext.bn_bwd_reduce_cl(x, dout, out, mean, istd, sums.reshape(-1),
                                  c, ctx.cnt, relu, parts)
+            apply_sums = sums
+            if getattr(ctx, "sync_world", 1) > 1:
+                apply_sums = _allreduce(sums.clone())
             ext.bn_bwd_apply_cl(x, dout, out, mean, istd, ctx.gflat,
-                                sums.reshape(-1), dx, c, ctx.cnt, relu,
-                                ctx.has_affine, ctx.use_batch, parts)
+                                apply_sums.reshape(-1), dx, c, ctx.cnt, relu,
+                                ctx.has_affine, ctx.use_batch, parts,
+                                cnt_total)
         else:
             for p in range(parts):
                 sl = slice(p * b, (p + 1) * b)
@@ -371,11 +438,17 @@ class _HipBatchNormMulti(torch.autograd.Function):
                                   mean[p * c:(p + 1) * c],
                                   istd[p * c:(p + 1) * c],
                                   sums[p].reshape(-1), relu)
+            apply_sums = sums
+            if getattr(ctx, "sync_world", 1) > 1:
+                apply_sums = _allreduce(sums.clone())
+            for p in range(parts):
+                sl = slice(p * b, (p + 1) * b)
                 ext.bn_bwd_apply(x[sl], dout[sl], out[sl],
                                  mean[p * c:(p + 1) * c],
                                  istd[p * c:(p + 1) * c],
-                                 ctx.gflat, sums[p].reshape(-1), dx[sl], relu,
-                                 ctx.has_affine, ctx.use_batch)
+                                 ctx.gflat, apply_sums[p].reshape(-1), dx[sl],
+                                 relu, ctx.has_affine, ctx.use_batch,
+                                 cnt_total)
 
         if ctx.has_affine:
             # dgamma = sum dy * xhat ; dbeta = sum dy  (sums[:,1] is dy*xhat)

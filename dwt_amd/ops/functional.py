@@ -333,11 +333,10 @@ def whiten_multi(
                training=training, mode=mode, relu=relu, ns_iters=ns_iters,
                track_running_stats=track_running_stats)
     if os.environ.get("DWT_AMD_STATS_SYNC") == "1":
-        # cross-rank batch statistics ('sync' stats mode) —
-        # currently served by the torch path (exact; HIP fast path is
-        # per-rank stats + parallel.sync_stats())
+        # cross-rank batch statistics ('sync' stats mode):
+        # on GPU the HIP kernels serve it with a partial-sums -> all-reduce
+        # -> finalize split; torch path elsewhere
         cfg["stats_sync"] = True
-        return WhitenMulti.apply(x, gamma, beta, running_means, running_vars, cfg)
     if x.is_cuda:
         from ..kernels import dispatch
         if dispatch.available():
@@ -490,7 +489,6 @@ def batch_norm_multi(
                relu=relu, track_running_stats=track_running_stats)
     if os.environ.get("DWT_AMD_STATS_SYNC") == "1":
         cfg["stats_sync"] = True
-        return BatchNormMulti.apply(x, gamma, beta, running_means, running_vars, cfg)
     if x.is_cuda:
         from ..kernels import dispatch
         if dispatch.available():
