@@ -31,11 +31,16 @@ def build(verbose: bool = True) -> str | None:
 
     from torch.utils.cpp_extension import load
 
+    # DWT_AMD_DEBUG=1: -O1 -g with device asserts enabled (see
+    # docs/SANITIZER.md) — pair with HIP_LAUNCH_BLOCKING=1 when hunting a
+    # corrupting launch
+    debug = os.environ.get("DWT_AMD_DEBUG") == "1"
+    opt = ["-O1", "-g"] if debug else ["-O3", "-DNDEBUG"]
     mod = load(
         name="_dwt_hip",
         sources=sources,
-        extra_cflags=["-O3", "-std=c++17"],
-        extra_cuda_cflags=["-O3", "-std=c++17"],
+        extra_cflags=opt + ["-std=c++17"],
+        extra_cuda_cflags=opt + ["-std=c++17"],
         build_directory=build_dir,
         verbose=verbose,
         is_python_module=False,

@@ -55,7 +55,8 @@ def build_args(argv=None):
     parser.add_argument('--group_size', type=int, default=4)
     parser.add_argument('--dtype', choices=['float32', 'bfloat16'],
                         default='bfloat16' if torch.cuda.is_available() else 'float32')
-    parser.add_argument('--whiten_mode', choices=['chol', 'zca'], default='chol')
+    parser.add_argument('--whiten_mode', choices=['chol', 'zca'], default='zca',
+                        help='ZCA Newton-Schulz (primary) or Cholesky (reference-parity mode)')
     parser.add_argument('--checkpoint_path', type=str, default='')
     parser.add_argument('--checkpoint_every', type=int, default=1000)
     parser.add_argument('--resume', action='store_true')

@@ -53,7 +53,8 @@ def build_args(argv=None):
     parser.add_argument('--synthetic_size', type=int, default=2048)
     parser.add_argument('--data_root', type=str, default='../data')
     parser.add_argument('--dtype', choices=['float32', 'bfloat16'], default='float32')
-    parser.add_argument('--whiten_mode', choices=['chol', 'zca'], default='chol')
+    parser.add_argument('--whiten_mode', choices=['chol', 'zca'], default='zca',
+                        help='ZCA Newton-Schulz (primary) or Cholesky (reference-parity mode)')
     parser.add_argument('--checkpoint_path', type=str, default='')
     parser.add_argument('--resume', action='store_true')
     parser.add_argument('--metrics_jsonl', type=str, default='')
