@@ -288,7 +288,7 @@ DEV_INLINE void write_tile(const StageRegs& rg, unsigned short* lds) {
 // (within-shape A/B on the R50 shapes).  SWZ: XCD-aware bijective block
 // remap (guide T1) — only when the grid has several N-tiles to share.
 template <int MODE, bool RELU, bool HAS_BIAS, bool PIPE, bool SWZ>
-__global__ __launch_bounds__(THREADS, 2) void conv_implicit_gemm_kernel(
+__global__ __launch_bounds__(THREADS, PIPE ? 2 : 3) void conv_implicit_gemm_kernel(
     const bf16* __restrict__ x, const bf16* __restrict__ wgt,
     const float* __restrict__ bias, bf16* __restrict__ out, ConvParams cp,
     int mtiles, int ntiles) {
@@ -619,7 +619,7 @@ struct DgradClsParams {
   int dp[16], dq[16], rsIdx[16];
 };
 
-__global__ __launch_bounds__(THREADS, 2) void dgrad_cls_kernel(
+__global__ __launch_bounds__(THREADS, 3) void dgrad_cls_kernel(
     const bf16* __restrict__ dy, const bf16* __restrict__ wd,
     bf16* __restrict__ dx, DgradClsParams gp, int mtiles, int ntiles) {
   __shared__ unsigned short lds_a[BM * LDS_PITCH];
