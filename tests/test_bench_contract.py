@@ -32,6 +32,23 @@ def test_bench_single_process_cpu():
     assert j["config"]["parallelism"] == "dp1"
 
 
+def test_bench_torchrun_ws8_cpu():
+    """The 8-GPU scaling-run shape (dp8) end to end on CPU/gloo: 8 ranks,
+    bucketed DP, max-over-ranks timing, one JSON line from rank 0."""
+    env = dict(os.environ)
+    env["MASTER_ADDR"] = "127.0.0.1"
+    r = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node", "8", "--master-addr", "127.0.0.1",
+         "--master-port", "29713", "bench.py", "--gpus", "8", "--steps", "1",
+         "--warmup", "0", "--device", "cpu", "--layers", "1,1,1,1"],
+        cwd=REPO, capture_output=True, text=True, timeout=1200, env=env)
+    assert r.returncode == 0, (r.stdout[-1000:], r.stderr[-2000:])
+    j = _last_json_line(r.stdout)
+    assert j["n_gpus"] == 8
+    assert j["config"]["parallelism"] == "dp8"
+
+
 def test_bench_torchrun_ws2_cpu():
     """The driver launches bench via torch.distributed.run for N>1; verify the
     whole path works (gloo on CPU, 127.0.0.1 rendezvous)."""
