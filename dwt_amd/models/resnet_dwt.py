@@ -167,7 +167,11 @@ class ResNetDWT(nn.Module):
         self.layer4 = self._make_layer(block, 512, layers[3], bn_dict, layer=4,
                                        group_size=group_size, stride=2)
         self.avgpool = nn.AdaptiveAvgPool2d((1, 1))
-        self.fc_out = nn.Linear(512 * block.expansion, num_classes)
+        if _conv_cls() is not nn.Conv2d:  # DWT_AMD_CONV=hip: fc on MFMA too
+            from ..ops.mfma import MFMALinear
+            self.fc_out = MFMALinear(512 * block.expansion, num_classes)
+        else:
+            self.fc_out = nn.Linear(512 * block.expansion, num_classes)
 
         for m in self.modules():
             if isinstance(m, nn.Conv2d):
