@@ -200,6 +200,7 @@ class _HipWhitenMulti(torch.autograd.Function):
         ctx.saved_mat = saved_mat
         ctx.has_affine = has_affine
         ctx.gflat = gflat
+        ctx.bflat = bflat
         ctx.save_for_backward(x, gamma, out)
         return out
 
@@ -232,7 +233,7 @@ class _HipWhitenMulti(torch.autograd.Function):
         dx = torch.empty_like(x)
 
         if layout == "cl":
-            ext.whiten_bwd_reduce_cl(x, dout, out, mean, wmat, ctx.gflat,
+            ext.whiten_bwd_reduce_cl(x, dout, mean, wmat, ctx.gflat, ctx.bflat,
                                      dW, dgb.reshape(-1), g, c, m_count,
                                      relu, ctx.has_affine, parts)
         else:
@@ -270,7 +271,7 @@ class _HipWhitenMulti(torch.autograd.Function):
             corr = torch.empty(0, device=dev)
 
         if layout == "cl":
-            ext.whiten_bwd_apply_cl(x, dout, out, mean, wmat, ctx.gflat,
+            ext.whiten_bwd_apply_cl(x, dout, mean, wmat, ctx.gflat, ctx.bflat,
                                     S, corr, dx, g, c, m_count, relu,
                                     ctx.has_affine, use_batch, parts)
         else:
@@ -398,6 +399,7 @@ class _HipBatchNormMulti(torch.autograd.Function):
         ctx.istd = istd
         ctx.has_affine = has_affine
         ctx.gflat = gflat
+        ctx.bflat = bflat
         ctx.use_batch = use_batch
         ctx.save_for_backward(x, gamma, out)
         return out
@@ -422,12 +424,13 @@ class _HipBatchNormMulti(torch.autograd.Function):
         dx = torch.empty_like(x)
         cnt_total = getattr(ctx, "cnt_total", ctx.cnt)
         if layout == "cl":
-            ext.bn_bwd_reduce_cl(x, dout, out, mean, istd, sums.reshape(-1),
-                                 c, ctx.cnt, relu, parts)
+            ext.bn_bwd_reduce_cl(x, dout, mean, istd, ctx.gflat, ctx.bflat,
+                                 sums.reshape(-1), c, ctx.cnt, relu,
+                                 ctx.has_affine, parts)
             apply_sums = sums
             if getattr(ctx, "sync_world", 1) > 1:
                 apply_sums = _allreduce(sums.clone())
-            ext.bn_bwd_apply_cl(x, dout, out, mean, istd, ctx.gflat,
+            ext.bn_bwd_apply_cl(x, dout, mean, istd, ctx.gflat, ctx.bflat,
                                 apply_sums.reshape(-1), dx, c, ctx.cnt, relu,
                                 ctx.has_affine, ctx.use_batch, parts,
                                 cnt_total)

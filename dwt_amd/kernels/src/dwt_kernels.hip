@@ -1178,13 +1178,14 @@ __global__ void whiten_apply_nhwc_kernel(
 template <typename T, int G>
 __global__ void whiten_bwd_reduce_nhwc_kernel(
     const T* __restrict__ x, const T* __restrict__ dout,
-    const T* __restrict__ out, const float* __restrict__ mean,
+    const float* __restrict__ mean,
     const float* __restrict__ W, const T* __restrict__ gamma,
+    const T* __restrict__ beta,
     float* __restrict__ dWacc, float* __restrict__ dgb,
     int C, int64_t M, int relu, int has_affine) {
   const int part = blockIdx.y;
   const int64_t poff = (int64_t)part * M * C;
-  x += poff; dout += poff; out += poff;
+  x += poff; dout += poff;
   mean += (int64_t)part * C;
   dgb += (int64_t)part * 2 * C;
   const int CW = C < 256 ? C : 256;
@@ -1194,11 +1195,12 @@ __global__ void whiten_bwd_reduce_nhwc_kernel(
   const int rows_per_iter = blockDim.x / GW;
   const int row_in_block = threadIdx.x / GW;
 
-  float m_[G], Wr[G][G], gm[G];
+  float m_[G], Wr[G][G], gm[G], bt[G];
 #pragma unroll
   for (int i = 0; i < G; ++i) {
     m_[i] = mean[c0 + i];
     gm[i] = has_affine ? ldf(gamma + c0 + i) : 1.f;
+    bt[i] = has_affine ? ldf(beta + c0 + i) : 0.f;
 #pragma unroll
     for (int j = 0; j < G; ++j) Wr[i][j] = W[((int64_t)grp * G + i) * G + j];
   }
@@ -1212,23 +1214,21 @@ __global__ void whiten_bwd_reduce_nhwc_kernel(
   const int64_t stride = (int64_t)gridDim.x * rows_per_iter;
   for (int64_t m = (int64_t)blockIdx.x * rows_per_iter + row_in_block; m < M;
        m += stride) {
-    float xv[G], dyv[G], ov[G];
+    float xv[G], dyv[G];
     load_group<T, G>(x + m * C + c0, xv);
     load_group<T, G>(dout + m * C + c0, dyv);
-    if (relu) {
-      load_group<T, G>(out + m * C + c0, ov);
-#pragma unroll
-      for (int j = 0; j < G; ++j) dyv[j] = ov[j] > 0.f ? dyv[j] : 0.f;
-    }
 #pragma unroll
     for (int j = 0; j < G; ++j) xv[j] -= m_[j];
 #pragma unroll
     for (int i = 0; i < G; ++i) {
-      const float dy = dyv[i];
-      db[i] += dy;
       float y0 = 0.f;
 #pragma unroll
       for (int j = 0; j < G; ++j) y0 += Wr[i][j] * xv[j];
+      // relu mask recomputed from the same fp32 pre-activation the forward
+      // stored (bf16 rounding preserves sign) — saves the out re-read
+      float dy = dyv[i];
+      if (relu && gm[i] * y0 + bt[i] <= 0.f) dy = 0.f;
+      db[i] += dy;
       dg[i] += dy * y0;
       const float dy0 = dy * gm[i];
 #pragma unroll
@@ -1265,14 +1265,15 @@ __global__ void whiten_bwd_reduce_nhwc_kernel(
 template <typename T, int G>
 __global__ void whiten_bwd_apply_nhwc_kernel(
     const T* __restrict__ x, const T* __restrict__ dout,
-    const T* __restrict__ out, const float* __restrict__ mean,
+    const float* __restrict__ mean,
     const float* __restrict__ W, const T* __restrict__ gamma,
+    const T* __restrict__ beta,
     const float* __restrict__ S, const float* __restrict__ corr,
     T* __restrict__ dx, int C, int64_t M, int relu, int has_affine,
     int train_stats) {
   const int part = blockIdx.y;
   const int64_t poff = (int64_t)part * M * C;
-  x += poff; dout += poff; out += poff; dx += poff;
+  x += poff; dout += poff; dx += poff;
   mean += (int64_t)part * C;
   corr += (int64_t)part * C;
   const int CW = C < 256 ? C : 256;
@@ -1282,14 +1283,16 @@ __global__ void whiten_bwd_apply_nhwc_kernel(
   const int rows_per_iter = blockDim.x / GW;
   const int row_in_block = threadIdx.x / GW;
 
-  float m_[G], Wt[G][G], Sr[G][G], gm[G], cr[G];
+  float m_[G], Wt[G][G], Wr[G][G], Sr[G][G], gm[G], bt[G], cr[G];
 #pragma unroll
   for (int i = 0; i < G; ++i) {
     m_[i] = mean[c0 + i];
     gm[i] = has_affine ? ldf(gamma + c0 + i) : 1.f;
+    bt[i] = has_affine ? ldf(beta + c0 + i) : 0.f;
     cr[i] = train_stats ? corr[c0 + i] : 0.f;
 #pragma unroll
     for (int j = 0; j < G; ++j) {
+      Wr[i][j] = W[((int64_t)grp * G + i) * G + j];
       Wt[i][j] = W[((int64_t)grp * G + j) * G + i];  // W^T
       Sr[i][j] = train_stats ? S[((int64_t)grp * G + i) * G + j] : 0.f;
     }
@@ -1297,19 +1300,22 @@ __global__ void whiten_bwd_apply_nhwc_kernel(
   const int64_t stride = (int64_t)gridDim.x * rows_per_iter;
   for (int64_t m = (int64_t)blockIdx.x * rows_per_iter + row_in_block; m < M;
        m += stride) {
-    float xv[G], dyv[G], ov[G], r[G];
+    float xv[G], dyv[G], r[G];
     load_group<T, G>(x + m * C + c0, xv);
     load_group<T, G>(dout + m * C + c0, dyv);
+#pragma unroll
+    for (int j = 0; j < G; ++j) xv[j] -= m_[j];
     if (relu) {
-      load_group<T, G>(out + m * C + c0, ov);
 #pragma unroll
-      for (int j = 0; j < G; ++j) dyv[j] = ov[j] > 0.f ? dyv[j] : 0.f;
+      for (int i = 0; i < G; ++i) {
+        float y0 = 0.f;
+#pragma unroll
+        for (int j = 0; j < G; ++j) y0 += Wr[i][j] * xv[j];
+        if (gm[i] * y0 + bt[i] <= 0.f) dyv[i] = 0.f;
+      }
     }
 #pragma unroll
-    for (int j = 0; j < G; ++j) {
-      xv[j] -= m_[j];
-      dyv[j] *= gm[j];
-    }
+    for (int j = 0; j < G; ++j) dyv[j] *= gm[j];
 #pragma unroll
     for (int i = 0; i < G; ++i) {
       float a = -cr[i];
@@ -1405,12 +1411,13 @@ __global__ void bn_apply_nhwc_kernel(
 template <typename T>
 __global__ void bn_bwd_reduce_nhwc_kernel(
     const T* __restrict__ x, const T* __restrict__ dout,
-    const T* __restrict__ out, const float* __restrict__ mean,
-    const float* __restrict__ istd, float* __restrict__ sums, int C,
-    int64_t M, int relu) {
+    const float* __restrict__ mean,
+    const float* __restrict__ istd, const T* __restrict__ gamma,
+    const T* __restrict__ beta, float* __restrict__ sums, int C,
+    int64_t M, int relu, int has_affine) {
   const int part = blockIdx.y;
   const int64_t poff = (int64_t)part * M * C;
-  x += poff; dout += poff; out += poff;
+  x += poff; dout += poff;
   mean += (int64_t)part * C;
   istd += (int64_t)part * C;
   sums += (int64_t)part * 2 * C;
@@ -1420,24 +1427,28 @@ __global__ void bn_bwd_reduce_nhwc_kernel(
   const int c0 = blockIdx.z * 1024 + (threadIdx.x % NCH) * VC;
   const int rows_per_iter = blockDim.x / NCH;
   const int row_in_block = threadIdx.x / NCH;
-  float s_dy[VC], s_dyxh[VC];
+  float s_dy[VC], s_dyxh[VC], mu[VC], is[VC], gm[VC], bt[VC];
 #pragma unroll
-  for (int k = 0; k < VC; ++k) { s_dy[k] = 0.f; s_dyxh[k] = 0.f; }
+  for (int k = 0; k < VC; ++k) {
+    s_dy[k] = 0.f; s_dyxh[k] = 0.f;
+    mu[k] = mean[c0 + k];
+    is[k] = istd[c0 + k];
+    gm[k] = has_affine ? ldf(gamma + c0 + k) : 1.f;
+    bt[k] = has_affine ? ldf(beta + c0 + k) : 0.f;
+  }
   const int64_t stride = (int64_t)gridDim.x * rows_per_iter;
   for (int64_t m = (int64_t)blockIdx.x * rows_per_iter + row_in_block; m < M;
        m += stride) {
-    float xv[VC], dv[VC], ov[VC];
+    float xv[VC], dv[VC];
     load_group<T, VC>(x + m * C + c0, xv);
     load_group<T, VC>(dout + m * C + c0, dv);
-    if (relu) {
-      load_group<T, VC>(out + m * C + c0, ov);
-#pragma unroll
-      for (int k = 0; k < VC; ++k) dv[k] = ov[k] > 0.f ? dv[k] : 0.f;
-    }
 #pragma unroll
     for (int k = 0; k < VC; ++k) {
+      const float xh = (xv[k] - mu[k]) * is[k];
+      // relu mask recomputed (sign survives the forward's bf16 store)
+      if (relu && gm[k] * xh + bt[k] <= 0.f) dv[k] = 0.f;
       s_dy[k] += dv[k];
-      s_dyxh[k] += dv[k] * (xv[k] - mean[c0 + k]) * istd[c0 + k];
+      s_dyxh[k] += dv[k] * xh;
     }
   }
   __shared__ float lacc[2 * 1024];
@@ -1459,13 +1470,14 @@ __global__ void bn_bwd_reduce_nhwc_kernel(
 template <typename T>
 __global__ void bn_bwd_apply_nhwc_kernel(
     const T* __restrict__ x, const T* __restrict__ dout,
-    const T* __restrict__ out, const float* __restrict__ mean,
+    const float* __restrict__ mean,
     const float* __restrict__ istd, const T* __restrict__ gamma,
+    const T* __restrict__ beta,
     const float* __restrict__ sums, T* __restrict__ dx, int C, int64_t M,
     float inv_m, int relu, int has_affine, int use_batch) {
   const int part = blockIdx.y;
   const int64_t poff = (int64_t)part * M * C;
-  x += poff; dout += poff; out += poff; dx += poff;
+  x += poff; dout += poff; dx += poff;
   mean += (int64_t)part * C;
   istd += (int64_t)part * C;
   sums += (int64_t)part * 2 * C;
@@ -1475,32 +1487,29 @@ __global__ void bn_bwd_apply_nhwc_kernel(
   const int c0 = blockIdx.z * 1024 + (threadIdx.x % NCH) * VC;
   const int rows_per_iter = blockDim.x / NCH;
   const int row_in_block = threadIdx.x / NCH;
-  float mu[VC], is[VC], gm[VC], mdy[VC], mdyxh[VC];
+  float mu[VC], is[VC], gm[VC], bt[VC], mdy[VC], mdyxh[VC];
 #pragma unroll
   for (int k = 0; k < VC; ++k) {
     const int c = c0 + k;
     mu[k] = mean[c];
     is[k] = istd[c];
     gm[k] = has_affine ? ldf(gamma + c) : 1.f;
+    bt[k] = has_affine ? ldf(beta + c) : 0.f;
     mdy[k] = use_batch ? sums[c] * inv_m : 0.f;
     mdyxh[k] = use_batch ? sums[C + c] * inv_m : 0.f;
   }
   const int64_t stride = (int64_t)gridDim.x * rows_per_iter;
   for (int64_t m = (int64_t)blockIdx.x * rows_per_iter + row_in_block; m < M;
        m += stride) {
-    float xv[VC], dv[VC], ov[VC];
+    float xv[VC], dv[VC];
     load_group<T, VC>(x + m * C + c0, xv);
     load_group<T, VC>(dout + m * C + c0, dv);
-    if (relu) {
-      load_group<T, VC>(out + m * C + c0, ov);
-#pragma unroll
-      for (int k = 0; k < VC; ++k) dv[k] = ov[k] > 0.f ? dv[k] : 0.f;
-    }
 #pragma unroll
     for (int k = 0; k < VC; ++k) {
+      const float xh = (xv[k] - mu[k]) * is[k];
+      if (relu && gm[k] * xh + bt[k] <= 0.f) dv[k] = 0.f;
       const float dxh = dv[k] * gm[k];
       if (use_batch) {
-        const float xh = (xv[k] - mu[k]) * is[k];
         dv[k] = (dxh - gm[k] * mdy[k] - xh * gm[k] * mdyxh[k]) * is[k];
       } else {
         dv[k] = dxh * is[k];
@@ -2298,8 +2307,9 @@ void whiten_apply_cl(Tensor x, Tensor mean, Tensor W, Tensor gamma, Tensor beta,
   });
 }
 
-void whiten_bwd_reduce_cl(Tensor x, Tensor dout, Tensor out, Tensor mean,
-                          Tensor W, Tensor gamma, Tensor dWacc, Tensor dgb,
+void whiten_bwd_reduce_cl(Tensor x, Tensor dout, Tensor mean,
+                          Tensor W, Tensor gamma, Tensor beta,
+                          Tensor dWacc, Tensor dgb,
                           int64_t g, int64_t C, int64_t M, bool relu,
                           bool has_affine, int64_t parts) {
   const int zslices = (C + 255) / 256;
@@ -2311,17 +2321,19 @@ void whiten_bwd_reduce_cl(Tensor x, Tensor dout, Tensor out, Tensor mean,
       hipLaunchKernelGGL((dwt::whiten_bwd_reduce_nhwc_kernel<scalar_t, G>),
                          grid, dim3(256), 0, cur_stream(),
                          x.data_ptr<scalar_t>(), dout.data_ptr<scalar_t>(),
-                         out.data_ptr<scalar_t>(), mean.data_ptr<float>(),
+                         mean.data_ptr<float>(),
                          W.data_ptr<float>(),
                          has_affine ? gamma.data_ptr<scalar_t>() : nullptr,
+                         has_affine ? beta.data_ptr<scalar_t>() : nullptr,
                          dWacc.data_ptr<float>(), dgb.data_ptr<float>(), (int)C,
                          M, relu ? 1 : 0, has_affine ? 1 : 0);
     });
   });
 }
 
-void whiten_bwd_apply_cl(Tensor x, Tensor dout, Tensor out, Tensor mean,
-                         Tensor W, Tensor gamma, Tensor S, Tensor corr,
+void whiten_bwd_apply_cl(Tensor x, Tensor dout, Tensor mean,
+                         Tensor W, Tensor gamma, Tensor beta,
+                         Tensor S, Tensor corr,
                          Tensor dx, int64_t g, int64_t C, int64_t M, bool relu,
                          bool has_affine, bool train_stats, int64_t parts) {
   const int zslices = (C + 255) / 256;
@@ -2332,9 +2344,10 @@ void whiten_bwd_apply_cl(Tensor x, Tensor dout, Tensor out, Tensor mean,
       hipLaunchKernelGGL((dwt::whiten_bwd_apply_nhwc_kernel<scalar_t, G>),
                          grid, dim3(256), 0, cur_stream(),
                          x.data_ptr<scalar_t>(),
-                         dout.data_ptr<scalar_t>(), out.data_ptr<scalar_t>(),
+                         dout.data_ptr<scalar_t>(),
                          mean.data_ptr<float>(), W.data_ptr<float>(),
                          has_affine ? gamma.data_ptr<scalar_t>() : nullptr,
+                         has_affine ? beta.data_ptr<scalar_t>() : nullptr,
                          S.data_ptr<float>(), corr.data_ptr<float>(),
                          dx.data_ptr<scalar_t>(), (int)C, M, relu ? 1 : 0,
                          has_affine ? 1 : 0, train_stats ? 1 : 0);
@@ -2390,9 +2403,10 @@ void bn_apply_cl(Tensor x, Tensor mean, Tensor istd, Tensor gamma, Tensor beta,
   });
 }
 
-void bn_bwd_reduce_cl(Tensor x, Tensor dout, Tensor out, Tensor mean,
-                      Tensor istd, Tensor sums, int64_t C, int64_t M,
-                      bool relu, int64_t parts) {
+void bn_bwd_reduce_cl(Tensor x, Tensor dout, Tensor mean,
+                      Tensor istd, Tensor gamma, Tensor beta,
+                      Tensor sums, int64_t C, int64_t M,
+                      bool relu, bool has_affine, int64_t parts) {
   const int zslices = (C + 1023) / 1024;
   DISPATCH_FT(x, "bn_bwd_reduce_cl", [&] {
     const int NCH = (C < 1024 ? C : 1024) / 4;
@@ -2400,14 +2414,18 @@ void bn_bwd_reduce_cl(Tensor x, Tensor dout, Tensor out, Tensor mean,
                                  zslices * parts), parts, zslices);
     hipLaunchKernelGGL((dwt::bn_bwd_reduce_nhwc_kernel<scalar_t>), grid,
                        dim3(256), 0, cur_stream(), x.data_ptr<scalar_t>(),
-                       dout.data_ptr<scalar_t>(), out.data_ptr<scalar_t>(),
+                       dout.data_ptr<scalar_t>(),
                        mean.data_ptr<float>(), istd.data_ptr<float>(),
-                       sums.data_ptr<float>(), (int)C, M, relu ? 1 : 0);
+                       has_affine ? gamma.data_ptr<scalar_t>() : nullptr,
+                       has_affine ? beta.data_ptr<scalar_t>() : nullptr,
+                       sums.data_ptr<float>(), (int)C, M, relu ? 1 : 0,
+                       has_affine ? 1 : 0);
   });
 }
 
-void bn_bwd_apply_cl(Tensor x, Tensor dout, Tensor out, Tensor mean,
-                     Tensor istd, Tensor gamma, Tensor sums, Tensor dx,
+void bn_bwd_apply_cl(Tensor x, Tensor dout, Tensor mean,
+                     Tensor istd, Tensor gamma, Tensor beta,
+                     Tensor sums, Tensor dx,
                      int64_t C, int64_t M, bool relu, bool has_affine,
                      bool use_batch, int64_t parts, int64_t count) {
   const int zslices = (C + 1023) / 1024;
@@ -2417,9 +2435,10 @@ void bn_bwd_apply_cl(Tensor x, Tensor dout, Tensor out, Tensor mean,
     dim3 grid(nhwc_elem_blocks(M, rpi, zslices * parts), parts, zslices);
     hipLaunchKernelGGL((dwt::bn_bwd_apply_nhwc_kernel<scalar_t>), grid,
                        dim3(256), 0, cur_stream(), x.data_ptr<scalar_t>(),
-                       dout.data_ptr<scalar_t>(), out.data_ptr<scalar_t>(),
+                       dout.data_ptr<scalar_t>(),
                        mean.data_ptr<float>(), istd.data_ptr<float>(),
                        has_affine ? gamma.data_ptr<scalar_t>() : nullptr,
+                       has_affine ? beta.data_ptr<scalar_t>() : nullptr,
                        sums.data_ptr<float>(), dx.data_ptr<scalar_t>(), (int)C,
                        M, 1.0f / (float)count, relu ? 1 : 0,
                        has_affine ? 1 : 0, use_batch ? 1 : 0);
