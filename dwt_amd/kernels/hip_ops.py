@@ -1,10 +1,11 @@
 """GPU autograd Functions backed by the gfx950 HIP kernels.
 
 Mirrors the algorithm contract of dwt_amd/ops/functional.py exactly (these
-are numerically tested against it in tests/test_gpu_kernels.py).  Fast path
-covers group sizes {2, 4, 8} — every config the reference models use with the
-default group_size=4 — and fp32/bf16 inputs; other group sizes fall back to
-the torch implementation (warned once).
+are numerically tested against it in tests/test_gpu_kernels.py).  The
+register-blocked fast path covers group sizes {2, 4, 8}; 8 < g <= 32 runs
+the generic LDS-tiled NCHW kernels (every group size the reference models
+can construct); fp32/bf16 inputs; anything else falls back to the torch
+implementation (warned once).
 """
 from __future__ import annotations
 
