@@ -212,6 +212,61 @@ def test_ddp_grad_parity_modes(world, dtype_name, reduce_fp32):
     assert results.get(timeout=5) == "ok"
 
 
+def _run_unused_param(rank, world, port, results):
+    """A parameter whose grad never materializes (unused head) must not hang
+    the bucket protocol: sync() launches the incomplete bucket with a zeroed
+    segment, and the reduced grads still average correctly."""
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    try:
+        from dwt_amd.parallel import BucketedDataParallel
+
+        class TwoHead(torch.nn.Module):
+            def __init__(self):
+                super().__init__()
+                self.body = torch.nn.Linear(16, 16)
+                self.used = torch.nn.Linear(16, 4)
+                self.unused = torch.nn.Linear(16, 4)
+
+            def forward(self, x):
+                return self.used(torch.relu(self.body(x)))
+
+        torch.manual_seed(0)
+        model = TwoHead()
+        ddp = BucketedDataParallel(model, bucket_cap_mb=1.0)  # one bucket
+        torch.manual_seed(50 + rank)
+        x = torch.randn(4, 16)
+        model(x).pow(2).mean().backward()
+        ddp.sync()  # must not deadlock on the never-fired hooks
+        g = model.body.weight.grad.clone()
+        avg = g.clone()
+        dist.all_reduce(avg)
+        # sync() already averaged: re-averaging the averaged grad across
+        # ranks must be a fixed point
+        assert torch.allclose(g, avg / world, atol=1e-6)
+        assert model.unused.weight.grad is not None  # view installed, zeros
+        assert model.unused.weight.grad.abs().max() == 0
+        if rank == 0:
+            results.put("ok")
+    finally:
+        dist.destroy_process_group()
+
+
+def test_ddp_unused_param_bucket():
+    port = 29951
+    ctx = mp.get_context("spawn")
+    results = ctx.Queue()
+    procs = [ctx.Process(target=_run_unused_param, args=(r, 2, port, results))
+             for r in range(2)]
+    for p in procs:
+        p.start()
+    for p in procs:
+        p.join(timeout=180)
+    assert all(p.exitcode == 0 for p in procs), [p.exitcode for p in procs]
+    assert results.get(timeout=5) == "ok"
+
+
 def _run_stats_sync_parity(rank, world, port, results):
     os.environ["MASTER_ADDR"] = "127.0.0.1"
     os.environ["MASTER_PORT"] = str(port)
