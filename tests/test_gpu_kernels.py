@@ -10,7 +10,6 @@ import torch
 
 pytestmark = pytest.mark.gpu
 
-requires_gpu = pytest.mark.skipif(not torch.cuda.is_available(), reason="needs ROCm GPU")
 
 
 @pytest.fixture(scope="module")
