@@ -1275,7 +1275,7 @@ __global__ void whiten_bwd_apply_nhwc_kernel(
   const int64_t poff = (int64_t)part * M * C;
   x += poff; dout += poff; dx += poff;
   mean += (int64_t)part * C;
-  corr += (int64_t)part * C;
+  if (train_stats) corr += (int64_t)part * C;  // empty tensor in eval
   const int CW = C < 256 ? C : 256;
   const int GW = CW / G;
   const int c0 = blockIdx.z * 256 + (threadIdx.x % GW) * G;
