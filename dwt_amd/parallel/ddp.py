@@ -100,6 +100,9 @@ class BucketedDataParallel:
 
         self.buckets: List[_Bucket] = []
         self._param_bucket = {}
+        # while True, grad-ready hooks do NOT launch reductions (gradient
+        # accumulation across several backwards; use accumulate()):
+        self.accumulating = False
         if self.enabled:
             self._build_buckets(bucket_cap_mb)
             self._install_hooks()
@@ -148,9 +151,30 @@ class BucketedDataParallel:
                 # freshly accumulated grad into the bucket and alias it
                 v.copy_(p.grad.detach().reshape(p.shape))
                 p.grad = v
+        if self.accumulating:
+            # a reduction launched now would race with the NEXT backward's
+            # accumulation into the same buffer — defer to the final one
+            return
         b.pending += 1
         if b.pending >= len(b.params):
             self._launch(b)
+
+    def accumulate(self):
+        """Context manager: suppress hook-launched reductions for backwards
+        whose gradients should accumulate locally (call ``sync()`` after the
+        final backward as usual)."""
+        import contextlib
+
+        @contextlib.contextmanager
+        def _ctx():
+            prev = self.accumulating
+            self.accumulating = True
+            try:
+                yield
+            finally:
+                self.accumulating = prev
+
+        return _ctx()
 
     def _launch(self, b: _Bucket):
         if b.work is not None:

@@ -212,6 +212,53 @@ def test_ddp_grad_parity_modes(world, dtype_name, reduce_fp32):
     assert results.get(timeout=5) == "ok"
 
 
+def _run_grad_accumulation(rank, world, port, results):
+    """Two accumulated backwards + one sync must equal the average of the
+    SUMMED local grads (accumulate() defers the hook launches so the async
+    reduce cannot race the second backward)."""
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    try:
+        from dwt_amd.parallel import BucketedDataParallel
+        torch.manual_seed(0)
+        model = torch.nn.Linear(8, 4)
+        clone = torch.nn.Linear(8, 4)
+        clone.load_state_dict(model.state_dict())
+        ddp = BucketedDataParallel(model, bucket_cap_mb=0.001)
+        torch.manual_seed(10 + rank)
+        xs = [torch.randn(4, 8) for _ in range(2)]
+        with ddp.accumulate():
+            model(xs[0]).pow(2).mean().backward()
+        model(xs[1]).pow(2).mean().backward()
+        ddp.sync()
+        for x in xs:
+            clone(x).pow(2).mean().backward()
+        for p, pc in zip(model.parameters(), clone.parameters()):
+            avg = pc.grad.clone()
+            dist.all_reduce(avg)
+            avg /= world
+            assert torch.allclose(p.grad, avg, atol=1e-6)
+        if rank == 0:
+            results.put("ok")
+    finally:
+        dist.destroy_process_group()
+
+
+def test_ddp_grad_accumulation():
+    port = 29971
+    ctx = mp.get_context("spawn")
+    results = ctx.Queue()
+    procs = [ctx.Process(target=_run_grad_accumulation, args=(r, 2, port, results))
+             for r in range(2)]
+    for p in procs:
+        p.start()
+    for p in procs:
+        p.join(timeout=180)
+    assert all(p.exitcode == 0 for p in procs), [p.exitcode for p in procs]
+    assert results.get(timeout=5) == "ok"
+
+
 def _run_unused_param(rank, world, port, results):
     """A parameter whose grad never materializes (unused head) must not hang
     the bucket protocol: sync() launches the incomplete bucket with a zeroed
