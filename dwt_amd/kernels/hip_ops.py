@@ -14,8 +14,8 @@ import torch
 
 from . import dispatch
 
-_FAST_G = (2, 4, 8)      # register-blocked kernels (NCHW + NHWC)
-_GEN_G_MAX = 32          # LDS-tiled generic kernels, NCHW (g=16/32 digits)
+_GEN_G_MAX = 32  # register-blocked kernels cover g in {2,4,8};
+                 # 8 < g <= 32 runs the LDS-tiled generic NCHW kernels
 _warned = set()
 
 
