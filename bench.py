@@ -52,6 +52,14 @@ def main():
     world = int(os.environ.get("WORLD_SIZE", "1"))
     rank = int(os.environ.get("RANK", "0"))
     local_rank = int(os.environ.get("LOCAL_RANK", rank))
+    if world > 1:
+        # per-rank MIOpen find/user DBs: N ranks finding the same conv
+        # shapes contend on the shared DB's file locks otherwise (the find
+        # runs inside the untimed warmup either way)
+        os.environ.setdefault("MIOPEN_USER_DB_PATH",
+                              f"/tmp/miopen-rank{local_rank}")
+        os.environ.setdefault("MIOPEN_CACHE_DIR",
+                              f"/tmp/miopen-cache-rank{local_rank}")
     use_cuda = torch.cuda.is_available() and args.device != "cpu"
     if args.device == "cpu":
         # keep the CPU smoke path fast
