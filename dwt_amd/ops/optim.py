@@ -26,7 +26,10 @@ def _hip_available(params):
     if not params or not params[0].is_cuda:
         return False
     from ..kernels import dispatch
-    return dispatch.available()
+    if not dispatch.available():
+        dispatch.require_or_warn("fused_optimizer")
+        return False
+    return True
 
 
 class _FusedBase(Optimizer):

@@ -21,7 +21,10 @@ def _hip_ok(x):
     if not x.is_cuda or x.dtype not in (torch.float32, torch.bfloat16):
         return False
     from ..kernels import dispatch
-    return dispatch.available()
+    if not dispatch.available():
+        dispatch.require_or_warn("pooling")  # no silent eager fallback on GPU
+        return False
+    return True
 
 
 class MaxPool2dFn(torch.autograd.Function):
