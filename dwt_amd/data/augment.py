@@ -45,3 +45,38 @@ def gaussian_blur(x: torch.Tensor, sigma: float = 0.1) -> torch.Tensor:
 # reference-style aliases
 _random_affine_augmentation = random_affine_augmentation
 _gaussian_blur = gaussian_blur
+
+
+# ---------------------------------------------------------------------------
+# Batched, device-agnostic version (per-image CPU augmentation
+# can starve 8 GPUs — this one runs the whole target batch on the GPU).
+# ---------------------------------------------------------------------------
+
+def affine_augment_batch(x: torch.Tensor,
+                         mats: torch.Tensor = None,
+                         std: float = 0.1,
+                         generator: torch.Generator = None) -> torch.Tensor:
+    """Batched equivalent of ``random_affine_augmentation``: x (N, C, H, W)
+    on any device; per-image 2x2 matrix = I + N(0, std) entries, applied
+    origin-anchored exactly like the scalar scipy path (out[o] = in[inv @ o]
+    in pixel index space, zeros outside).  ``mats`` overrides the random
+    matrices (N, 2, 2), acting on (row, col) indices like the scipy path."""
+    n, _, h, w = x.shape
+    dev = x.device
+    if mats is None:
+        mats = torch.eye(2, device=dev).expand(n, 2, 2).clone()
+        mats = mats + std * torch.randn(n, 2, 2, device=dev,
+                                        generator=generator)
+    inv = torch.linalg.inv(mats.to(torch.float64)).to(torch.float32)  # (N,2,2)
+    ys = torch.arange(h, device=dev, dtype=torch.float32)
+    xs = torch.arange(w, device=dev, dtype=torch.float32)
+    gy, gx = torch.meshgrid(ys, xs, indexing="ij")       # (H, W) row/col
+    o = torch.stack([gy, gx])                            # (2, H, W)
+    # p = inv @ o per image: (N, 2, H, W)
+    p = torch.einsum("nij,jhw->nihw", inv, o)
+    grid = torch.empty(n, h, w, 2, device=dev, dtype=torch.float32)
+    grid[..., 0] = 2.0 * p[:, 1] / max(w - 1, 1) - 1.0   # x (col)
+    grid[..., 1] = 2.0 * p[:, 0] / max(h - 1, 1) - 1.0   # y (row)
+    return torch.nn.functional.grid_sample(
+        x.float(), grid, mode="bilinear", padding_mode="zeros",
+        align_corners=True).to(x.dtype)
