@@ -80,3 +80,16 @@ def affine_augment_batch(x: torch.Tensor,
     return torch.nn.functional.grid_sample(
         x.float(), grid, mode="bilinear", padding_mode="zeros",
         align_corners=True).to(x.dtype)
+
+
+def gpu_target_views(x: torch.Tensor, std: float = 0.1, flip_p: float = 0.5,
+                     generator: torch.Generator = None) -> torch.Tensor:
+    """Build the MEC duplicate target view on-device: random horizontal flip
+    + the random affine (the reference's blur is a no-op at its sigma).
+    Opt-in (--gpu_augment): the affine runs AFTER normalization here, so
+    out-of-frame pixels are 0 in normalized space rather than the
+    reference's normalized-zero constant — a border-only deviation."""
+    n = x.shape[0]
+    flip = torch.rand(n, device=x.device, generator=generator) < flip_p
+    out = torch.where(flip[:, None, None, None], torch.flip(x, dims=[-1]), x)
+    return affine_augment_batch(out, std=std, generator=generator)

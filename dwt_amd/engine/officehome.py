@@ -85,13 +85,25 @@ def _train_loop(args, model, device, source_iter, target_iter,
             source_iter = iter(source_train_loader)
             source_data, source_y = next(source_iter)
         try:
-            target_data, target_data_dup, _ = next(target_iter)
+            tb = next(target_iter)
         except StopIteration:
             target_iter = iter(target_train_loader)
-            target_data, target_data_dup, _ = next(target_iter)
+            tb = next(target_iter)
 
-        data = torch.cat((source_data, target_data, target_data_dup), dim=0)
-        data = data.to(device, non_blocking=True).to(dtype)
+        gpu_aug = getattr(args, "gpu_augment", False)
+        if gpu_aug:
+            # loader yields (data, label); the duplicate MEC view is built
+            # on-device (batched flip + affine — data/augment.py)
+            from ..data import augment
+            target_data = tb[0]
+            pair = torch.cat((source_data, target_data), dim=0)
+            pair = pair.to(device, non_blocking=True).to(dtype)
+            tdev = pair[source_data.shape[0]:]
+            data = torch.cat((pair, augment.gpu_target_views(tdev)), dim=0)
+        else:
+            target_data, target_data_dup, _ = tb
+            data = torch.cat((source_data, target_data, target_data_dup), dim=0)
+            data = data.to(device, non_blocking=True).to(dtype)
         source_y = source_y.to(device, non_blocking=True)
 
         optimizer.zero_grad(set_to_none=True)

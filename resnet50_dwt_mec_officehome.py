@@ -66,15 +66,21 @@ def build_args(argv=None):
                              "cross-rank synced batch stats")
     parser.add_argument('--stats_passes', type=int, default=10,
                         help='target-stats re-estimation passes before final test')
+    parser.add_argument('--gpu_augment', action='store_true',
+                        help='build the duplicate MEC target view on-device '
+                             '(batched flip+affine) instead of in the CPU '
+                             'loader workers — keeps 8-GPU runs unstarved')
     return parser.parse_args(argv)
 
 
 def build_loaders(args, rank, world):
+    aug_in_loader = not getattr(args, 'gpu_augment', False)
     if args.synthetic:
         src = SyntheticOfficeHome(args.synthetic_size, args.num_classes,
                                   args.img_crop_size, seed=1)
         tgt = SyntheticOfficeHome(args.synthetic_size, args.num_classes,
-                                  args.img_crop_size, transform_aug=True, seed=2)
+                                  args.img_crop_size,
+                                  transform_aug=aug_in_loader, seed=2)
         tgt_test = SyntheticOfficeHome(max(args.synthetic_size // 4, 64),
                                        args.num_classes, args.img_crop_size, seed=2)
     else:
@@ -95,7 +101,8 @@ def build_loaders(args, rank, world):
         ])
         src = ImageFolder(root=args.s_dset_path, transform=data_transform)
         tgt = ImageFolder(root=args.t_dset_path, transform=data_transform,
-                          transform_aug=data_transform_dup)
+                          transform_aug=data_transform_dup if aug_in_loader
+                          else None)
         tgt_test = ImageFolder(root=args.t_dset_path, transform=data_transform)
 
     samplers = {}

@@ -212,3 +212,20 @@ def test_emergency_checkpoint_on_failure(tmp_path):
     model2 = ResNetDWT(Bottleneck, [1, 1, 1, 1], None, num_classes=5)
     it = ckptmod.load_training_state(path + ".emergency", model2)
     assert it == 2  # failed during iteration 2's forward
+
+
+def test_officehome_entrypoint_gpu_augment_cpu():
+    """--gpu_augment: the duplicate MEC view is built on-device by the
+    engine (batched flip+affine) instead of in the loader workers."""
+    import subprocess, sys, os as _os
+    repo = _os.path.dirname(_os.path.dirname(_os.path.abspath(__file__)))
+    r = subprocess.run(
+        [sys.executable, "resnet50_dwt_mec_officehome.py", "--synthetic",
+         "--synthetic_size", "64", "--num_iters", "3", "--check_acc_step",
+         "100", "--source_batch_size", "4", "--test_batch_size", "8",
+         "--num_workers", "0", "--log_interval", "1", "--img_crop_size",
+         "64", "--gpu_augment", "--stats_passes", "1"],
+        cwd=repo, capture_output=True, text=True, timeout=900)
+    assert r.returncode == 0, r.stderr[-2000:]
+    assert "Train Iter" in r.stdout
+    assert "Test set" in r.stdout
