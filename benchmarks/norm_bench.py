@@ -70,7 +70,9 @@ def main():
         t_b = timeit(bwd, iters=10, warmup=3)
 
         fwd_traffic = 3 * nbytes            # stats read + apply read + write
-        bwd_traffic = 7 * nbytes            # reduce 3 reads + apply 3 reads 1 write
+        # reduce reads x,dy; apply reads x,dy + writes dx (the ReLU mask is
+        # recomputed — the stored out is no longer re-read in backward)
+        bwd_traffic = 5 * nbytes
         lines.append(f"| {kind} C={c} {s}x{s} B={3*b} | {t_f*1e3:.3f} | "
                      f"{fwd_traffic/t_f/1e12:.2f} | {t_b*1e3:.3f} | "
                      f"{bwd_traffic/t_b/1e12:.2f} |")
